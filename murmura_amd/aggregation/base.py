@@ -1,0 +1,140 @@
+"""Aggregator base + flat-state helpers.
+
+Reference surface: murmura/aggregation/base.py:9-170. The reference aggregates
+CPU state dicts key-by-key with ``.item()`` host syncs per tensor; here every
+algorithm operates on flat state vectors ([P]) and stacked neighbor states
+([k, P]) so the hot math is a handful of single-kernel launches (SURVEY.md
+§2.9) and stays entirely on-device.
+
+Statistics discipline: per-round scalars are kept as 0-dim device tensors in
+Python lists and only materialized to floats inside ``get_statistics()`` —
+the round loop never forces a host sync for bookkeeping.
+"""
+
+from __future__ import annotations
+
+import abc
+from typing import Any, Dict, List, Optional
+
+import torch
+from torch import Tensor
+
+from murmura_amd import ops
+
+
+class EvalContext:
+    """What loss-eval-style aggregators (UBAR, EvidentialTrust, DMTT scoring)
+    need to score a foreign model state: a scratch model bound to a flat buffer
+    (cheap pointer rebind instead of the reference's deepcopy-per-neighbor,
+    evidential_trust.py:237) plus local data."""
+
+    def __init__(self, store, loader, device, evidential: bool = False):
+        self.store = store  # FlatParamStore scratch
+        self.loader = loader
+        self.device = device
+        self.evidential = evidential
+        self._batch_iter = None
+
+    def next_batch(self):
+        """One training batch, cycling through the loader."""
+        if self._batch_iter is None:
+            self._batch_iter = iter(self.loader)
+        try:
+            batch = next(self._batch_iter)
+        except StopIteration:
+            self._batch_iter = iter(self.loader)
+            batch = next(self._batch_iter)
+        x, y = batch
+        return x.to(self.device), y.to(self.device)
+
+    @torch.no_grad()
+    def loss_on_batch(self, flat_state: Tensor, batch) -> Tensor:
+        """CE loss of ``flat_state`` on one batch (UBAR stage 2)."""
+        x, y = batch
+        self.store.copy_from_flat(flat_state)
+        self.store.model.eval()
+        logits = self.store.model(x.to(dtype=self.store.dtype))
+        loss_sum, _ = ops.ce_loss_acc(logits, y)
+        return loss_sum / max(1, x.shape[0])
+
+    @torch.no_grad()
+    def evidential_score(self, flat_state: Tensor, max_samples: int = 100):
+        """(mean vacuity, accuracy) of a foreign evidential model on local
+        data (EvidentialTrust / DMTT model scoring)."""
+        self.store.copy_from_flat(flat_state)
+        self.store.model.eval()
+        seen = 0
+        vac_sum = torch.zeros((), device=self.device)
+        correct = torch.zeros((), device=self.device)
+        for x, y in self.loader:
+            if seen >= max_samples:
+                break
+            x = x.to(self.device, dtype=self.store.dtype)
+            y = y.to(self.device)
+            if seen + x.shape[0] > max_samples:
+                take = max_samples - seen
+                x, y = x[:take], y[:take]
+            logits = self.store.model(x)
+            v, _, _, c = ops.evidential_stats(logits, y)
+            vac_sum = vac_sum + v
+            correct = correct + c
+            seen += x.shape[0]
+        n = max(1, seen)
+        return vac_sum / n, correct.float() / n
+
+
+class Aggregator(abc.ABC):
+    """Base aggregator. Subclasses implement ``aggregate``; every node gets
+    its own stateful instance (history/EMA is per-node, like the reference's
+    per-node aggregator factory, utils/factories.py:83-88)."""
+
+    requires_eval_context: bool = False
+
+    @abc.abstractmethod
+    def aggregate(
+        self,
+        node_id: int,
+        own_state: Tensor,
+        neighbor_states: Tensor,
+        round_num: int = 0,
+        **ctx: Any,
+    ) -> Tensor:
+        """Combine own flat state [P] with stacked neighbor states [k, P];
+        returns the new flat state [P]. Must not mutate inputs."""
+
+    def get_statistics(self) -> Dict[str, Any]:
+        return {}
+
+
+def _to_float_list(vals: List[Tensor]) -> List[float]:
+    if not vals:
+        return []
+    return torch.stack([v.detach().float().cpu() for v in vals]).tolist()
+
+
+def blend(own: Tensor, neighbor_states: Tensor, neighbor_weights: Tensor, alpha: float) -> Tensor:
+    """alpha * own + (1 - alpha) * sum_i w_i * neighbor_i — the common final
+    step of BALANCE/UBAR/Sketchguard (reference: balance.py:140-175 etc.),
+    as ONE fused weighted-sum launch over [k+1, P]."""
+    stacked = torch.cat([own.unsqueeze(0), neighbor_states], dim=0)
+    w = torch.cat(
+        [
+            torch.full((1,), alpha, device=own.device, dtype=torch.float32),
+            (1.0 - alpha) * neighbor_weights.float(),
+        ]
+    )
+    return ops.weighted_sum(stacked, w)
+
+
+def accept_weights(
+    accept_mask: Tensor, dists: Tensor, min_neighbors: int = 1
+) -> Tensor:
+    """Normalized neighbor weights from a boolean accept mask, with the
+    reference's fallback: if fewer than ``min_neighbors`` accepted, accept the
+    closest neighbor instead (balance.py:133-135). Fully device-side."""
+    cnt = accept_mask.sum()
+    fallback = torch.zeros_like(dists)
+    fallback[torch.argmin(dists)] = 1.0
+    w = accept_mask.float() / cnt.clamp(min=1).float()
+    use_fb = (cnt < min_neighbors).float()
+    return use_fb * fallback + (1.0 - use_fb) * w

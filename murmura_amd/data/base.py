@@ -1,0 +1,18 @@
+"""Dataset protocol (reference: murmura/data/base.py:8-39)."""
+
+from __future__ import annotations
+
+from typing import List, Protocol, runtime_checkable
+
+from torch.utils.data import Dataset
+
+
+@runtime_checkable
+class DatasetProtocol(Protocol):
+    """Structural protocol for per-client dataset providers."""
+
+    def get_client_data(self, client_id: int) -> Dataset: ...
+
+    def get_num_clients(self) -> int: ...
+
+    def get_client_partitions(self) -> List[List[int]]: ...

@@ -1,0 +1,144 @@
+"""Evidential deep learning: heads, loss, and the wearables classifiers
+(reference: murmura/examples/wearables/models.py:18-347).
+
+Models output raw evidence logits; alpha = softplus(logits) + 1 is taken in
+the loss and in the fused evidential-stats kernel (K8), keeping the forward
+pass a plain MLP.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+from torch import Tensor, nn
+import torch.nn.functional as F
+
+
+class EvidentialHead(nn.Module):
+    """Final linear layer producing evidence logits
+    (reference: models.py:18-46)."""
+
+    def __init__(self, in_features: int, num_classes: int):
+        super().__init__()
+        self.linear = nn.Linear(in_features, num_classes)
+        self.num_classes = num_classes
+
+    def forward(self, x: Tensor) -> Tensor:
+        return self.linear(x)
+
+
+def compute_uncertainty(logits: Tensor) -> dict:
+    """Dirichlet uncertainty measures from evidence logits
+    (reference: models.py:49-86)."""
+    alpha = F.softplus(logits.float()) + 1.0
+    s = alpha.sum(dim=1, keepdim=True)
+    k = logits.shape[1]
+    p = alpha / s
+    return {
+        "alpha": alpha,
+        "strength": s.squeeze(1),
+        "vacuity": k / s.squeeze(1),
+        "entropy": -(p * p.clamp_min(1e-10).log()).sum(dim=1),
+        "probs": p,
+    }
+
+
+class EvidentialLoss(nn.Module):
+    """EDL loss: one-hot MSE under Dirichlet expectation + annealed
+    KL(Dir(alpha_tilde) || Dir(1)) (reference: models.py:118-179).
+
+    kl_weight ramps linearly to ``max_kl_weight`` over ``annealing_rounds``.
+    """
+
+    def __init__(self, num_classes: int, annealing_rounds: int = 10, max_kl_weight: float = 0.1):
+        super().__init__()
+        self.num_classes = num_classes
+        self.annealing_rounds = max(1, int(annealing_rounds))
+        self.max_kl_weight = float(max_kl_weight)
+
+    def forward(self, logits: Tensor, targets: Tensor, round_num: int = 0) -> Tensor:
+        logits = logits.float()
+        alpha = F.softplus(logits) + 1.0
+        s = alpha.sum(dim=1, keepdim=True)
+        p = alpha / s
+        y = F.one_hot(targets, self.num_classes).float()
+        # expected MSE under Dirichlet: (y - p)^2 + p(1-p)/(S+1)
+        mse = ((y - p) ** 2).sum(dim=1) + (p * (1.0 - p) / (s + 1.0)).sum(dim=1)
+        # KL(Dir(alpha_tilde) || Dir(1)) on the misleading evidence
+        alpha_t = y + (1.0 - y) * alpha
+        kl = self._kl_to_uniform(alpha_t)
+        lam = self.max_kl_weight * min(1.0, round_num / self.annealing_rounds)
+        return (mse + lam * kl).mean()
+
+    def _kl_to_uniform(self, alpha: Tensor) -> Tensor:
+        k = float(self.num_classes)
+        s = alpha.sum(dim=1)
+        ln_b = torch.lgamma(alpha).sum(dim=1) - torch.lgamma(s)
+        ln_b_uni = torch.lgamma(torch.tensor(k, device=alpha.device))  # lgamma(K)·(-1) trick: ln B(1)=  -lgamma(K)
+        dg_s = torch.digamma(s).unsqueeze(1)
+        term = ((alpha - 1.0) * (torch.digamma(alpha) - dg_s)).sum(dim=1)
+        return ln_b_uni - ln_b + term
+
+
+def _mlp_block(in_f: int, out_f: int, dropout: float = 0.3) -> nn.Sequential:
+    return nn.Sequential(
+        nn.Linear(in_f, out_f),
+        nn.BatchNorm1d(out_f),
+        nn.ReLU(),
+        nn.Dropout(dropout),
+    )
+
+
+class EvidentialHARClassifier(nn.Module):
+    """UCI HAR: MLP 561-256-128-EDL6 (reference: models.py:187-243)."""
+
+    def __init__(self, in_features: int = 561, num_classes: int = 6):
+        super().__init__()
+        self.body = nn.Sequential(
+            _mlp_block(in_features, 256), _mlp_block(256, 128)
+        )
+        self.head = EvidentialHead(128, num_classes)
+
+    def forward(self, x: Tensor) -> Tensor:
+        return self.head(self.body(x.flatten(1)))
+
+
+class EvidentialPAMAP2Classifier(nn.Module):
+    """PAMAP2: windows 100x40 -> 4000-512-256-128-EDL12
+    (reference: models.py:246-286)."""
+
+    def __init__(self, in_features: int = 4000, num_classes: int = 12):
+        super().__init__()
+        self.body = nn.Sequential(
+            _mlp_block(in_features, 512), _mlp_block(512, 256), _mlp_block(256, 128)
+        )
+        self.head = EvidentialHead(128, num_classes)
+
+    def forward(self, x: Tensor) -> Tensor:
+        return self.head(self.body(x.flatten(1)))
+
+
+class EvidentialPPGDaLiAClassifier(nn.Module):
+    """PPG-DaLiA: windows 32x6 -> 192-256-128-64-EDL7
+    (reference: models.py:289-347)."""
+
+    def __init__(self, in_features: int = 192, num_classes: int = 7):
+        super().__init__()
+        self.body = nn.Sequential(
+            _mlp_block(in_features, 256), _mlp_block(256, 128), _mlp_block(128, 64)
+        )
+        self.head = EvidentialHead(64, num_classes)
+
+    def forward(self, x: Tensor) -> Tensor:
+        return self.head(self.body(x.flatten(1)))
+
+
+def get_evidential_loss(
+    num_classes: int, total_rounds: int = 50, max_kl_weight: float = 0.1
+) -> EvidentialLoss:
+    """Annealing horizon = rounds/2, lambda = 0.1, mirroring the reference's
+    build_criterion coupling (utils/factories.py:106-120)."""
+    return EvidentialLoss(
+        num_classes, annealing_rounds=max(1, total_rounds // 2), max_kl_weight=max_kl_weight
+    )

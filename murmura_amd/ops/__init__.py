@@ -1,0 +1,147 @@
+"""Op dispatch: hand-written CDNA4 HIP kernels on GPU, torch reference on CPU.
+
+Policy (per the build rules): when a tensor is on a ROCm device the native
+extension MUST be present and is always used — a silent eager fallback on the
+GPU box is a bug, so we raise instead. The torch implementations in
+``reference.py`` serve the CPU simulation backend and the numerics tests.
+
+Set MURMURA_DISABLE_NATIVE=1 to force the torch path everywhere (debug only).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional, Tuple
+
+import torch
+from torch import Tensor
+
+from murmura_amd.ops import reference as ref
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from murmura_amd.ops import _hip_loader
+
+        _EXT = _hip_loader.load()
+    except Exception as e:  # noqa: BLE001
+        _EXT_ERR = f"{type(e).__name__}: {e}"
+        _EXT = None
+    return _EXT
+
+
+def native_available() -> bool:
+    return _load_ext() is not None
+
+
+def _use_native(t: Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    if os.environ.get("MURMURA_DISABLE_NATIVE") == "1":
+        return False
+    ext = _load_ext()
+    if ext is None:
+        raise RuntimeError(
+            "murmura_amd HIP extension is required for GPU tensors but failed to "
+            f"load: {_EXT_ERR}. Build it with `python -m murmura_amd.ops.build` "
+            "(or __graft_entry__.build())."
+        )
+    return True
+
+
+# ------------------------------------------------------------------ K1
+def weighted_sum(stacked: Tensor, weights: Tensor, out: Optional[Tensor] = None) -> Tensor:
+    if _use_native(stacked):
+        return _EXT.weighted_sum(stacked, weights.to(stacked.device), out)
+    return ref.weighted_sum(stacked, weights, out)
+
+
+# ------------------------------------------------------------------ K2/K12
+def pairwise_sq_dists(stacked: Tensor) -> Tensor:
+    if _use_native(stacked):
+        return _EXT.pairwise_sq_dists(stacked)
+    return ref.pairwise_sq_dists(stacked)
+
+
+def pairwise_l2(stacked: Tensor) -> Tensor:
+    return pairwise_sq_dists(stacked).clamp_min(0).sqrt()
+
+
+def row_norms(stacked: Tensor) -> Tensor:
+    if _use_native(stacked):
+        return _EXT.row_norms(stacked.view(1, -1) if stacked.dim() == 1 else stacked).view(
+            () if stacked.dim() == 1 else (-1,)
+        )
+    return ref.row_norms(stacked)
+
+
+def l2_dists_to(own: Tensor, stacked: Tensor) -> Tensor:
+    if _use_native(stacked):
+        return _EXT.l2_dists_to(own, stacked)
+    return ref.l2_dists_to(own, stacked)
+
+
+# ------------------------------------------------------------------ K3
+def krum_scores(d2: Tensor, num_compromised: int) -> Tensor:
+    # m x m is tiny (m <= nodes); torch path is fine on both devices.
+    return ref.krum_scores(d2, num_compromised)
+
+
+def krum_select(d2: Tensor, num_compromised: int) -> Tensor:
+    return ref.krum_select(d2, num_compromised)
+
+
+# ------------------------------------------------------------------ K4/K5
+def count_sketch(stacked: Tensor, hash_idx: Tensor, signs: Tensor, sketch_size: int) -> Tensor:
+    if _use_native(stacked if stacked.dim() == 2 else stacked.view(1, -1)):
+        x = stacked if stacked.dim() == 2 else stacked.view(1, -1)
+        out = _EXT.count_sketch(x, hash_idx, signs, sketch_size)
+        return out if stacked.dim() == 2 else out.view(-1)
+    return ref.count_sketch(stacked, hash_idx, signs, sketch_size)
+
+
+make_sketch_tables = ref.make_sketch_tables
+
+
+# ------------------------------------------------------------------ K6
+def sgd_step(flat_params: Tensor, grad: Tensor, lr: float) -> None:
+    if _use_native(flat_params):
+        _EXT.sgd_step(flat_params, grad, lr)
+        return
+    ref.sgd_step(flat_params, grad, lr)
+
+
+# ------------------------------------------------------------------ K7
+def ce_loss_acc(logits: Tensor, targets: Tensor) -> Tuple[Tensor, Tensor]:
+    if _use_native(logits):
+        out = _EXT.ce_loss_acc(logits, targets)
+        return out[0], out[1]
+    return ref.ce_loss_acc(logits, targets)
+
+
+# ------------------------------------------------------------------ K8
+def evidential_stats(logits: Tensor, targets: Tensor):
+    if _use_native(logits):
+        out = _EXT.evidential_stats(logits, targets)
+        return out[0], out[1], out[2], out[3]
+    return ref.evidential_stats(logits, targets)
+
+
+# ------------------------------------------------------------------ K10
+def gaussian_inject(flat: Tensor, noise_std: float, seed: int, offset: int = 0) -> Tensor:
+    if _use_native(flat):
+        return _EXT.gaussian_inject(flat, noise_std, seed, offset)
+    return ref.gaussian_inject(flat, noise_std, seed, offset)
+
+
+# ------------------------------------------------------------------ K11
+def scale_inject(flat: Tensor, lam: float) -> Tensor:
+    if _use_native(flat):
+        return _EXT.scale_inject(flat, lam)
+    return ref.scale_inject(flat, lam)

