@@ -1,0 +1,197 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: FL rounds/sec, one node per MI355X GPU.
+
+Default config = BASELINE.json config 2: N-node fully-connected FedAvg with a
+ResNet-18-sized CNN in bf16 on synthetic 3x32x32 shards (random-init weights —
+no network for datasets). Weak scaling: each GPU is one FL node with a fixed
+local shard, so per-GPU work is constant as N grows.
+
+One step = one FL round = local epoch(s) of fused-SGD training + neighbor
+state exchange over RCCL/xGMI + aggregation + local evaluation (the
+reference's default round evaluates every round, core/network.py:80-94 —
+nothing is skipped inside the timed region).
+
+Usage:
+  python bench.py                          # N=1, quick
+  python bench.py --gpus N --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Other configs (BASELINE.json configs 3/4):
+  --algo krum --attack gaussian            # Krum under 20% Gaussian attackers
+  --algo sketchguard --model femnist-xlarge
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--algo", default="fedavg",
+                   choices=["fedavg", "krum", "balance", "sketchguard", "ubar",
+                            "evidential_trust"])
+    p.add_argument("--topology", default=None,
+                   help="default: fully (fedavg) / k-regular (krum)")
+    p.add_argument("--attack", default="none", choices=["none", "gaussian", "directed"])
+    p.add_argument("--model", default="resnet18",
+                   choices=["resnet18", "femnist-baseline", "femnist-xlarge", "mlp"])
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--shard", type=int, default=2048, help="samples per node")
+    p.add_argument("--batch-size", type=int, default=64)
+    p.add_argument("--local-epochs", type=int, default=1)
+    p.add_argument("--no-eval", action="store_true",
+                   help="skip per-round evaluation inside the timed region")
+    return p.parse_args()
+
+
+def build_config(args, world):
+    from murmura_amd.config.schema import Config
+
+    model_cfgs = {
+        "resnet18": ("models.resnet18", {"num_classes": 10},
+                     {"image_shape": [3, 32, 32], "num_classes": 10}),
+        "femnist-baseline": ("models.femnist", {"hidden": 2048},
+                             {"image_shape": [1, 28, 28], "num_classes": 62}),
+        "femnist-xlarge": ("models.femnist", {"hidden": 8192},
+                           {"image_shape": [1, 28, 28], "num_classes": 62}),
+        "mlp": ("models.mlp", {"in_features": 32, "hidden": 64, "num_classes": 4},
+                {"num_features": 32, "num_classes": 4}),
+    }
+    factory, mparams, dparams = model_cfgs[args.model]
+    topo = args.topology or ("k-regular" if args.algo == "krum" else "fully")
+    attack_cfg = {"enabled": False}
+    if args.attack != "none":
+        attack_cfg = {
+            "enabled": True,
+            "type": "gaussian" if args.attack == "gaussian" else "directed_deviation",
+            "percentage": 0.2,
+            "params": {"noise_std": 10.0},
+        }
+    agg_params = {}
+    if args.algo == "krum":
+        agg_params = {"num_compromised": max(1, int(0.2 * world))}
+    return Config(**{
+        "experiment": {"name": f"bench-{args.algo}", "seed": 42,
+                       "rounds": args.steps + args.warmup, "verbose": False},
+        "topology": {"type": topo, "num_nodes": world, "k": 4},
+        "aggregation": {"algorithm": args.algo, "params": agg_params},
+        "attack": attack_cfg,
+        "training": {"local_epochs": args.local_epochs,
+                     "batch_size": args.batch_size, "lr": 0.01},
+        "data": {"adapter": "synthetic",
+                 "params": {"num_samples": args.shard * world,
+                            "partition": "iid", **dparams}},
+        "model": {"factory": factory, "params": mparams},
+        "backend": "rccl",
+        "compute": {"dtype": args.dtype, "native_kernels": True},
+    })
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", args.gpus))
+    rank = int(os.environ.get("RANK", 0))
+    launched_distributed = "WORLD_SIZE" in os.environ
+
+    if world > 1 and not launched_distributed:
+        print("for N>1 launch via torch.distributed.run", file=sys.stderr)
+        sys.exit(2)
+
+    config = build_config(args, world)
+    use_cuda = torch.cuda.is_available()
+
+    import torch.distributed as dist
+
+    from murmura_amd.parallel.node_process import FLRoundLoop, init_distributed
+
+    if launched_distributed or world > 1:
+        device = init_distributed(config, rank, world)
+    else:
+        # single-process single-node path: no process group needed
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29531")
+        dist.init_process_group(
+            backend="nccl" if use_cuda else "gloo", rank=0, world_size=1
+        )
+        device = torch.device("cuda:0" if use_cuda else "cpu")
+        if use_cuda:
+            torch.cuda.set_device(device)
+
+    loop = FLRoundLoop(config, rank, world, device)
+
+    def one_round(r):
+        loop.run_round(r)
+        if not args.no_eval:
+            loop.evaluate_round(r)
+
+    def sync():
+        if use_cuda:
+            torch.cuda.synchronize()
+        dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    # warmup
+    for r in range(args.warmup):
+        one_round(r)
+    sync()
+    t0 = time.perf_counter()
+    for r in range(args.warmup, args.warmup + args.steps):
+        one_round(r)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    el = torch.tensor([elapsed], dtype=torch.float64)
+    if use_cuda:
+        el = el.to(device)
+    dist.all_reduce(el, op=dist.ReduceOp.MAX)
+    elapsed = float(el.cpu().item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    rounds_per_sec = args.steps / elapsed
+
+    if rank == 0:
+        result = {
+            "metric": "FL rounds/sec",
+            "value": rounds_per_sec,
+            "unit": "rounds/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "algorithm": args.algo,
+                "topology": config.topology.type,
+                "attack": args.attack,
+                "global_batch": args.batch_size * world,
+                "shard_per_node": args.shard,
+                "local_epochs": args.local_epochs,
+                "eval_every_round": not args.no_eval,
+                "parallelism": f"fl-node-per-gpu x{world} (rccl/xGMI)",
+            },
+        }
+        print(json.dumps(result))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -203,6 +203,13 @@ class Network:
         attack = build_attack(config)
         dtype = torch.bfloat16 if config.compute.dtype == "bf16" else torch.float32
 
+        def seeded_model(i: int) -> torch.nn.Module:
+            # per-node deterministic init: identical weights in the simulation
+            # and rccl backends regardless of construction order
+            with torch.random.fork_rng(devices=[]):
+                torch.manual_seed(config.experiment.seed + i)
+                return model_factory()
+
         nodes: List[Node] = []
         for i in range(n):
             dev = device if device is not None else get_device(config.compute.device, i)
@@ -221,7 +228,7 @@ class Network:
             nodes.append(
                 Node(
                     node_id=i,
-                    model=model_factory(),
+                    model=seeded_model(i),
                     train_loader=train_loader,
                     test_loader=test_loader,
                     aggregator=aggregator_factory(i),

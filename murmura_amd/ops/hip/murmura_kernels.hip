@@ -1,0 +1,691 @@
+// murmura_amd HIP/CDNA4 kernels (gfx950 / MI355X).
+//
+// Kernel inventory follows SURVEY.md §2.9 (K1-K12): the aggregation and
+// round-loop hot paths of a decentralized-FL engine, operating on flat
+// parameter vectors [P] and stacked states [m, P] (m = 1 + #neighbors,
+// typically <= 16 on one 8xMI355X box; P up to ~100M).
+//
+// Every kernel here is HBM-bandwidth-bound streaming work; the design rules
+// applied (per the CDNA4 programming guide):
+//  - wave64 everywhere; blocks of 256 threads; grids sized >> 256 workgroups
+//    via grid-stride loops so all 8 XCDs fill.
+//  - 16-byte packed loads per lane (float4 / 8x bf16) — the compiler does not
+//    reliably auto-vectorize bf16 element loads.
+//  - fp32 accumulation regardless of storage dtype.
+//  - cross-lane reductions in-register via __shfl_xor (64-lane), one LDS
+//    accumulator per block, one global atomic per block — no per-pair host
+//    syncs (the reference's aggregation does m^2 x num_keys .item() calls).
+//  - pairwise-L2 reads each row ONCE (register-tiled Gram accumulation,
+//    compile-time m) instead of re-reading per pair.
+//  - Philox4x32-10 counter RNG for on-GPU attack injection (deterministic
+//    per (seed, offset, index) with no generator state).
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <cstdint>
+
+#define WAVE 64
+#define BLOCK 256
+
+// ------------------------------------------------------------------ dtypes
+struct bf16raw {
+  unsigned short v;
+};
+
+__device__ __forceinline__ float bf2f(unsigned short h) {
+  unsigned int u = ((unsigned int)h) << 16;
+  return __uint_as_float(u);
+}
+
+__device__ __forceinline__ unsigned short f2bf(float f) {
+  unsigned int u = __float_as_uint(f);
+  if ((u & 0x7fffffffu) > 0x7f800000u) return (unsigned short)0x7fc0;  // NaN
+  unsigned int lsb = (u >> 16) & 1u;
+  u += 0x7fffu + lsb;  // round-to-nearest-even
+  return (unsigned short)(u >> 16);
+}
+
+__device__ __forceinline__ float to_f(float x) { return x; }
+__device__ __forceinline__ float to_f(bf16raw x) { return bf2f(x.v); }
+__device__ __forceinline__ void from_f(float& d, float s) { d = s; }
+__device__ __forceinline__ void from_f(bf16raw& d, float s) { d.v = f2bf(s); }
+
+// 16-byte pack: 4 x fp32 or 8 x bf16 per lane per load
+template <typename T>
+struct Pack16 {
+  static constexpr int N = 16 / sizeof(T);
+  T e[N];
+};
+
+static inline int grid_for(int64_t work_items, int per_block, int cap = 4096) {
+  int64_t b = (work_items + per_block - 1) / per_block;
+  if (b < 1) b = 1;
+  if (b > cap) b = cap;
+  return (int)b;
+}
+
+// wave-level fp32 sum over all 64 lanes
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+  return v;
+}
+
+// ================================================================== K1
+// out[p] = sum_i w[i] * x[i*P + p]
+template <typename T>
+__global__ void weighted_sum_kernel(const T* __restrict__ x,
+                                    const float* __restrict__ w, T* __restrict__ out,
+                                    int m, int64_t P) {
+  __shared__ float ws[256];
+  for (int i = threadIdx.x; i < m; i += blockDim.x) ws[i] = w[i];
+  __syncthreads();
+
+  constexpr int N = Pack16<T>::N;
+  const int64_t nvec = P / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    float acc[N];
+#pragma unroll
+    for (int k = 0; k < N; ++k) acc[k] = 0.0f;
+    for (int i = 0; i < m; ++i) {
+      Pack16<T> pk = reinterpret_cast<const Pack16<T>*>(x + (int64_t)i * P)[v];
+      const float wi = ws[i];
+#pragma unroll
+      for (int k = 0; k < N; ++k) acc[k] = fmaf(wi, to_f(pk.e[k]), acc[k]);
+    }
+    Pack16<T> po;
+#pragma unroll
+    for (int k = 0; k < N; ++k) from_f(po.e[k], acc[k]);
+    reinterpret_cast<Pack16<T>*>(out)[v] = po;
+  }
+  // scalar tail
+  for (int64_t p = nvec * N + blockIdx.x * blockDim.x + threadIdx.x; p < P; p += stride) {
+    float acc = 0.0f;
+    for (int i = 0; i < m; ++i) acc = fmaf(ws[i], to_f(x[(int64_t)i * P + p]), acc);
+    from_f(out[p], acc);
+  }
+}
+
+// ================================================================== K2
+// Gram matrix over rows: gram[i, j] = <x_i, x_j> for i <= j, each row read
+// exactly once (register-tiled over compile-time M).
+template <typename T, int M>
+__global__ void gram_kernel(const T* __restrict__ x, int64_t P, float* __restrict__ gram) {
+  constexpr int NPAIR = M * (M + 1) / 2;
+  constexpr int N = Pack16<T>::N;
+  __shared__ float lacc[NPAIR];
+  for (int t = threadIdx.x; t < NPAIR; t += blockDim.x) lacc[t] = 0.0f;
+  __syncthreads();
+
+  float acc[NPAIR];
+#pragma unroll
+  for (int t = 0; t < NPAIR; ++t) acc[t] = 0.0f;
+
+  const int64_t nvec = P / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    Pack16<T> rows[M];
+#pragma unroll
+    for (int i = 0; i < M; ++i) rows[i] = reinterpret_cast<const Pack16<T>*>(x + (int64_t)i * P)[v];
+    int t = 0;
+#pragma unroll
+    for (int i = 0; i < M; ++i) {
+#pragma unroll
+      for (int j = i; j < M; ++j) {
+        float s = acc[t];
+#pragma unroll
+        for (int k = 0; k < N; ++k) s = fmaf(to_f(rows[i].e[k]), to_f(rows[j].e[k]), s);
+        acc[t] = s;
+        ++t;
+      }
+    }
+  }
+  // scalar tail
+  for (int64_t p = nvec * N + blockIdx.x * blockDim.x + threadIdx.x; p < P; p += stride) {
+    float rowsf[M];
+#pragma unroll
+    for (int i = 0; i < M; ++i) rowsf[i] = to_f(x[(int64_t)i * P + p]);
+    int t = 0;
+#pragma unroll
+    for (int i = 0; i < M; ++i)
+#pragma unroll
+      for (int j = i; j < M; ++j) acc[t++] += rowsf[i] * rowsf[j];
+  }
+
+  // reduce: wave shuffle -> LDS -> one global atomic per pair per block
+#pragma unroll
+  for (int t = 0; t < NPAIR; ++t) {
+    float s = wave_sum(acc[t]);
+    if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(&lacc[t], s);
+  }
+  __syncthreads();
+  for (int t = threadIdx.x; t < NPAIR; t += blockDim.x) atomicAdd(&gram[t], lacc[t]);
+}
+
+// ================================================================== K12
+// out[i] += sum_p x[i,p]^2  (per-row squared norms)
+template <typename T>
+__global__ void row_sqnorm_kernel(const T* __restrict__ x, int64_t P, float* __restrict__ out) {
+  constexpr int N = Pack16<T>::N;
+  const T* row = x + (int64_t)blockIdx.y * P;
+  const int64_t nvec = P / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float acc = 0.0f;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    Pack16<T> pk = reinterpret_cast<const Pack16<T>*>(row)[v];
+#pragma unroll
+    for (int k = 0; k < N; ++k) {
+      float f = to_f(pk.e[k]);
+      acc = fmaf(f, f, acc);
+    }
+  }
+  for (int64_t p = nvec * N + blockIdx.x * blockDim.x + threadIdx.x; p < P; p += stride) {
+    float f = to_f(row[p]);
+    acc = fmaf(f, f, acc);
+  }
+  __shared__ float lsum;
+  if (threadIdx.x == 0) lsum = 0.0f;
+  __syncthreads();
+  float s = wave_sum(acc);
+  if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(&lsum, s);
+  __syncthreads();
+  if (threadIdx.x == 0) atomicAdd(&out[blockIdx.y], lsum);
+}
+
+// sq dists of k rows against one "own" row (fallback for k+1 > 16)
+template <typename T>
+__global__ void sqdist_to_kernel(const T* __restrict__ own, const T* __restrict__ x,
+                                 int64_t P, float* __restrict__ out) {
+  constexpr int N = Pack16<T>::N;
+  const T* row = x + (int64_t)blockIdx.y * P;
+  const int64_t nvec = P / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float acc = 0.0f;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    Pack16<T> a = reinterpret_cast<const Pack16<T>*>(row)[v];
+    Pack16<T> b = reinterpret_cast<const Pack16<T>*>(own)[v];
+#pragma unroll
+    for (int k = 0; k < N; ++k) {
+      float d = to_f(a.e[k]) - to_f(b.e[k]);
+      acc = fmaf(d, d, acc);
+    }
+  }
+  for (int64_t p = nvec * N + blockIdx.x * blockDim.x + threadIdx.x; p < P; p += stride) {
+    float d = to_f(row[p]) - to_f(own[p]);
+    acc = fmaf(d, d, acc);
+  }
+  __shared__ float lsum;
+  if (threadIdx.x == 0) lsum = 0.0f;
+  __syncthreads();
+  float s = wave_sum(acc);
+  if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(&lsum, s);
+  __syncthreads();
+  if (threadIdx.x == 0) atomicAdd(&out[blockIdx.y], lsum);
+}
+
+// ================================================================== K4
+// Count-Sketch: out[row, h[p]] += sign[p] * x[row, p].
+// LDS-privatized histogram per block (sketch fits LDS: default 1000 fp32),
+// one global atomic per bin per block.
+template <typename T>
+__global__ void count_sketch_kernel(const T* __restrict__ x, const int* __restrict__ h,
+                                    const float* __restrict__ sg, int64_t P, int S,
+                                    float* __restrict__ out) {
+  extern __shared__ float hist[];
+  for (int b = threadIdx.x; b < S; b += blockDim.x) hist[b] = 0.0f;
+  __syncthreads();
+
+  const T* row = x + (int64_t)blockIdx.y * P;
+  float* orow = out + (int64_t)blockIdx.y * S;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < P; p += stride) {
+    atomicAdd(&hist[h[p]], sg[p] * to_f(row[p]));
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < S; b += blockDim.x) {
+    if (hist[b] != 0.0f) atomicAdd(&orow[b], hist[b]);
+  }
+}
+
+// ================================================================== K6
+// p <- p - lr * g over the whole param prefix, one launch
+template <typename T>
+__global__ void sgd_step_kernel(T* __restrict__ p, const T* __restrict__ g, float lr,
+                                int64_t P) {
+  constexpr int N = Pack16<T>::N;
+  const int64_t nvec = P / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    Pack16<T> pv = reinterpret_cast<Pack16<T>*>(p)[v];
+    Pack16<T> gv = reinterpret_cast<const Pack16<T>*>(g)[v];
+#pragma unroll
+    for (int k = 0; k < N; ++k) from_f(pv.e[k], fmaf(-lr, to_f(gv.e[k]), to_f(pv.e[k])));
+    reinterpret_cast<Pack16<T>*>(p)[v] = pv;
+  }
+  for (int64_t i = nvec * N + blockIdx.x * blockDim.x + threadIdx.x; i < P; i += stride) {
+    from_f(p[i], fmaf(-lr, to_f(g[i]), to_f(p[i])));
+  }
+}
+
+// ================================================================== K10
+// Philox4x32-10 counter RNG + Box-Muller: out = x + N(0, sigma^2).
+__device__ __forceinline__ void philox_round(unsigned int& c0, unsigned int& c1,
+                                             unsigned int& c2, unsigned int& c3,
+                                             unsigned int k0, unsigned int k1) {
+  const unsigned int M0 = 0xD2511F53u, M1 = 0xCD9E8D57u;
+  unsigned long long p0 = (unsigned long long)M0 * c0;
+  unsigned long long p1 = (unsigned long long)M1 * c2;
+  unsigned int h0 = (unsigned int)(p0 >> 32), l0 = (unsigned int)p0;
+  unsigned int h1 = (unsigned int)(p1 >> 32), l1 = (unsigned int)p1;
+  unsigned int n0 = h1 ^ c1 ^ k0;
+  unsigned int n1 = l1;
+  unsigned int n2 = h0 ^ c3 ^ k1;
+  unsigned int n3 = l0;
+  c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+}
+
+__device__ __forceinline__ void philox4(unsigned long long seed, unsigned long long ctr_hi,
+                                        unsigned long long ctr_lo, unsigned int out[4]) {
+  unsigned int k0 = (unsigned int)seed, k1 = (unsigned int)(seed >> 32);
+  unsigned int c0 = (unsigned int)ctr_lo, c1 = (unsigned int)(ctr_lo >> 32);
+  unsigned int c2 = (unsigned int)ctr_hi, c3 = (unsigned int)(ctr_hi >> 32);
+  const unsigned int W0 = 0x9E3779B9u, W1 = 0xBB67AE85u;
+#pragma unroll
+  for (int r = 0; r < 10; ++r) {
+    philox_round(c0, c1, c2, c3, k0, k1);
+    k0 += W0; k1 += W1;
+  }
+  out[0] = c0; out[1] = c1; out[2] = c2; out[3] = c3;
+}
+
+__device__ __forceinline__ float u32_to_uniform(unsigned int u) {
+  // (0, 1]: (u + 1) * 2^-32
+  return ((float)u + 1.0f) * 2.3283064365386963e-10f;
+}
+
+template <typename T>
+__global__ void gaussian_inject_kernel(const T* __restrict__ x, T* __restrict__ out,
+                                       float sigma, unsigned long long seed,
+                                       unsigned long long offset, int64_t P) {
+  const int64_t nquad = (P + 3) / 4;  // 4 outputs per philox call
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t q = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; q < nquad; q += stride) {
+    unsigned int r[4];
+    philox4(seed, offset, (unsigned long long)q, r);
+    float u0 = u32_to_uniform(r[0]), u1 = u32_to_uniform(r[1]);
+    float u2 = u32_to_uniform(r[2]), u3 = u32_to_uniform(r[3]);
+    float r0 = sqrtf(-2.0f * __logf(u0)), a0;
+    float s0, c0v;
+    __sincosf(6.2831853071795864f * u1, &s0, &c0v);
+    float r1 = sqrtf(-2.0f * __logf(u2)), s1, c1v;
+    __sincosf(6.2831853071795864f * u3, &s1, &c1v);
+    float n[4] = {r0 * c0v, r0 * s0, r1 * c1v, r1 * s1};
+    a0 = sigma;
+    int64_t base = q * 4;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      int64_t i = base + k;
+      if (i < P) from_f(out[i], fmaf(a0, n[k], to_f(x[i])));
+    }
+  }
+}
+
+// ================================================================== K11
+template <typename T>
+__global__ void scale_kernel(const T* __restrict__ x, T* __restrict__ out, float lam,
+                             int64_t P) {
+  constexpr int N = Pack16<T>::N;
+  const int64_t nvec = P / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    Pack16<T> pv = reinterpret_cast<const Pack16<T>*>(x)[v];
+#pragma unroll
+    for (int k = 0; k < N; ++k) from_f(pv.e[k], lam * to_f(pv.e[k]));
+    reinterpret_cast<Pack16<T>*>(out)[v] = pv;
+  }
+  for (int64_t i = nvec * N + blockIdx.x * blockDim.x + threadIdx.x; i < P; i += stride) {
+    from_f(out[i], lam * to_f(x[i]));
+  }
+}
+
+// ================================================================== K7
+// Fused CE-loss + accuracy epilogue: one WAVE per row.
+// out[0] += sum_rows (logsumexp - logit[target]); out[1] += #correct
+template <typename T>
+__global__ void ce_loss_acc_kernel(const T* __restrict__ logits,
+                                   const int64_t* __restrict__ targets, int B, int C,
+                                   float* __restrict__ out) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;          // wave within block (4)
+  const int waves_per_block = blockDim.x / WAVE;
+  float loss_acc = 0.0f;
+  float corr_acc = 0.0f;
+  for (int row = blockIdx.x * waves_per_block + wid; row < B;
+       row += gridDim.x * waves_per_block) {
+    const T* lr = logits + (int64_t)row * C;
+    const int tgt = (int)targets[row];
+    float vmax = -3.4e38f;
+    int imax = 0;
+    float xt = 0.0f;
+    for (int c = lane; c < C; c += WAVE) {
+      float v = to_f(lr[c]);
+      if (v > vmax) { vmax = v; imax = c; }
+      if (c == tgt) xt = v;
+    }
+    // wave argmax (max value, then lowest index among ties)
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      float ov = __shfl_xor(vmax, off, WAVE);
+      int oi = __shfl_xor(imax, off, WAVE);
+      if (ov > vmax || (ov == vmax && oi < imax)) { vmax = ov; imax = oi; }
+    }
+    float se = 0.0f;
+    for (int c = lane; c < C; c += WAVE) se += __expf(to_f(lr[c]) - vmax);
+    se = wave_sum(se);
+    // broadcast xt from the lane that owns column tgt (c == tgt happens on
+    // lane tgt % WAVE only; other lanes hold 0)
+    float xt_b = __shfl(xt, tgt % WAVE, WAVE);
+    if (lane == 0) {
+      loss_acc += __logf(se) + vmax - xt_b;
+      corr_acc += (imax == tgt) ? 1.0f : 0.0f;
+    }
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0) {
+    atomicAdd(&out[0], loss_acc);
+    atomicAdd(&out[1], corr_acc);
+  }
+}
+
+// ================================================================== K8
+// Evidential stats: alpha = softplus(x) + 1; per-row vacuity K/S, entropy of
+// alpha/S, strength S, correctness; sums atomically accumulated.
+__device__ __forceinline__ float softplusf(float x) {
+  // numerically-stable log(1 + exp(x))
+  if (x > 20.0f) return x;
+  if (x < -20.0f) return __expf(x);
+  return __logf(1.0f + __expf(x));
+}
+
+template <typename T>
+__global__ void evidential_stats_kernel(const T* __restrict__ logits,
+                                        const int64_t* __restrict__ targets, int B, int C,
+                                        float* __restrict__ out) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  float vac_acc = 0.0f, ent_acc = 0.0f, str_acc = 0.0f, corr_acc = 0.0f;
+  for (int row = blockIdx.x * waves_per_block + wid; row < B;
+       row += gridDim.x * waves_per_block) {
+    const T* lr = logits + (int64_t)row * C;
+    const int tgt = (int)targets[row];
+    float ssum = 0.0f, amax = -3.4e38f;
+    int imax = 0;
+    for (int c = lane; c < C; c += WAVE) {
+      float a = softplusf(to_f(lr[c])) + 1.0f;
+      ssum += a;
+      if (a > amax) { amax = a; imax = c; }
+    }
+    ssum = wave_sum(ssum);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      float ov = __shfl_xor(amax, off, WAVE);
+      int oi = __shfl_xor(imax, off, WAVE);
+      if (ov > amax || (ov == amax && oi < imax)) { amax = ov; imax = oi; }
+    }
+    float ent = 0.0f;
+    for (int c = lane; c < C; c += WAVE) {
+      float a = softplusf(to_f(lr[c])) + 1.0f;
+      float p = a / ssum;
+      float pl = p > 1e-10f ? p : 1e-10f;
+      ent -= p * __logf(pl);
+    }
+    ent = wave_sum(ent);
+    if (lane == 0) {
+      vac_acc += (float)C / ssum;
+      ent_acc += ent;
+      str_acc += ssum;
+      corr_acc += (imax == tgt) ? 1.0f : 0.0f;
+    }
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0) {
+    atomicAdd(&out[0], vac_acc);
+    atomicAdd(&out[1], ent_acc);
+    atomicAdd(&out[2], str_acc);
+    atomicAdd(&out[3], corr_acc);
+  }
+}
+
+// =================================================================
+// bindings
+// =================================================================
+
+namespace {
+
+using torch::Tensor;
+
+inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_flat(const Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == at::kFloat || t.scalar_type() == at::kBFloat16,
+              name, " must be fp32 or bf16");
+}
+
+#define DISPATCH_FT(TENSOR, ...)                        \
+  if ((TENSOR).scalar_type() == at::kFloat) {            \
+    using elem_t = float;                                \
+    __VA_ARGS__;                                         \
+  } else {                                               \
+    using elem_t = bf16raw;                              \
+    __VA_ARGS__;                                         \
+  }
+
+Tensor weighted_sum(Tensor stacked, Tensor w, c10::optional<Tensor> out_opt) {
+  check_flat(stacked, "stacked");
+  TORCH_CHECK(stacked.dim() == 2, "stacked must be [m, P]");
+  int m = (int)stacked.size(0);
+  int64_t P = stacked.size(1);
+  TORCH_CHECK(m <= 256, "weighted_sum: m must be <= 256");
+  Tensor wf = w.to(stacked.device(), at::kFloat).contiguous();
+  Tensor out = out_opt.has_value() ? *out_opt : at::empty({P}, stacked.options());
+  TORCH_CHECK(out.is_contiguous() && out.numel() == P);
+  int blocks = grid_for(P / 4, BLOCK);
+  DISPATCH_FT(stacked, {
+    weighted_sum_kernel<elem_t><<<blocks, BLOCK, 0, cur_stream()>>>(
+        (const elem_t*)stacked.data_ptr(), wf.data_ptr<float>(),
+        (elem_t*)out.data_ptr(), m, P);
+  });
+  return out;
+}
+
+template <typename elem_t, int M>
+void launch_gram(const Tensor& x, Tensor& gram_flat, int64_t P) {
+  int blocks = grid_for(P / 4, BLOCK, 2048);
+  gram_kernel<elem_t, M><<<blocks, BLOCK, 0, cur_stream()>>>(
+      (const elem_t*)x.data_ptr(), P, gram_flat.data_ptr<float>());
+}
+
+Tensor pairwise_sq_dists(Tensor stacked) {
+  check_flat(stacked, "stacked");
+  TORCH_CHECK(stacked.dim() == 2);
+  int m = (int)stacked.size(0);
+  int64_t P = stacked.size(1);
+  if (m > 16) {
+    // plain library GEMM path for large m
+    Tensor x = stacked.to(at::kFloat);
+    Tensor g = at::matmul(x, x.t());
+    Tensor sq = g.diagonal();
+    return (sq.unsqueeze(0) + sq.unsqueeze(1) - 2.0 * g).clamp_min_(0.0);
+  }
+  int npair = m * (m + 1) / 2;
+  Tensor gram_flat = at::zeros({npair}, stacked.options().dtype(at::kFloat));
+  DISPATCH_FT(stacked, {
+    switch (m) {
+      case 1: launch_gram<elem_t, 1>(stacked, gram_flat, P); break;
+      case 2: launch_gram<elem_t, 2>(stacked, gram_flat, P); break;
+      case 3: launch_gram<elem_t, 3>(stacked, gram_flat, P); break;
+      case 4: launch_gram<elem_t, 4>(stacked, gram_flat, P); break;
+      case 5: launch_gram<elem_t, 5>(stacked, gram_flat, P); break;
+      case 6: launch_gram<elem_t, 6>(stacked, gram_flat, P); break;
+      case 7: launch_gram<elem_t, 7>(stacked, gram_flat, P); break;
+      case 8: launch_gram<elem_t, 8>(stacked, gram_flat, P); break;
+      case 9: launch_gram<elem_t, 9>(stacked, gram_flat, P); break;
+      case 10: launch_gram<elem_t, 10>(stacked, gram_flat, P); break;
+      case 11: launch_gram<elem_t, 11>(stacked, gram_flat, P); break;
+      case 12: launch_gram<elem_t, 12>(stacked, gram_flat, P); break;
+      case 13: launch_gram<elem_t, 13>(stacked, gram_flat, P); break;
+      case 14: launch_gram<elem_t, 14>(stacked, gram_flat, P); break;
+      case 15: launch_gram<elem_t, 15>(stacked, gram_flat, P); break;
+      case 16: launch_gram<elem_t, 16>(stacked, gram_flat, P); break;
+    }
+  });
+  // unpack upper-triangular flat gram -> full [m, m] sq-dist matrix (tiny)
+  Tensor gram = at::zeros({m, m}, gram_flat.options());
+  auto idx = at::triu_indices(m, m, 0, gram_flat.options().dtype(at::kLong));
+  gram.index_put_({idx[0], idx[1]}, gram_flat);
+  gram = gram + gram.t() - at::diag(gram.diagonal());
+  Tensor sq = gram.diagonal();
+  return (sq.unsqueeze(0) + sq.unsqueeze(1) - 2.0 * gram).clamp_min_(0.0);
+}
+
+Tensor row_norms(Tensor stacked) {
+  check_flat(stacked, "stacked");
+  TORCH_CHECK(stacked.dim() == 2);
+  int m = (int)stacked.size(0);
+  int64_t P = stacked.size(1);
+  Tensor out = at::zeros({m}, stacked.options().dtype(at::kFloat));
+  dim3 grid(grid_for(P / 4, BLOCK, 1024), m);
+  DISPATCH_FT(stacked, {
+    row_sqnorm_kernel<elem_t><<<grid, BLOCK, 0, cur_stream()>>>(
+        (const elem_t*)stacked.data_ptr(), P, out.data_ptr<float>());
+  });
+  return out.sqrt_();
+}
+
+Tensor l2_dists_to(Tensor own, Tensor stacked) {
+  check_flat(stacked, "stacked");
+  check_flat(own, "own");
+  TORCH_CHECK(stacked.dim() == 2 && own.numel() == stacked.size(1));
+  TORCH_CHECK(own.scalar_type() == stacked.scalar_type());
+  int k = (int)stacked.size(0);
+  int64_t P = stacked.size(1);
+  Tensor out = at::zeros({k}, stacked.options().dtype(at::kFloat));
+  dim3 grid(grid_for(P / 4, BLOCK, 1024), k);
+  DISPATCH_FT(stacked, {
+    sqdist_to_kernel<elem_t><<<grid, BLOCK, 0, cur_stream()>>>(
+        (const elem_t*)own.data_ptr(), (const elem_t*)stacked.data_ptr(), P,
+        out.data_ptr<float>());
+  });
+  return out.clamp_min_(0.0).sqrt_();
+}
+
+Tensor count_sketch(Tensor stacked, Tensor h, Tensor sg, int64_t S) {
+  check_flat(stacked, "stacked");
+  TORCH_CHECK(stacked.dim() == 2);
+  int m = (int)stacked.size(0);
+  int64_t P = stacked.size(1);
+  TORCH_CHECK(S * sizeof(float) <= 160 * 1024 - 1024, "sketch too large for LDS");
+  Tensor hi = h.to(stacked.device(), at::kInt).contiguous();
+  Tensor sf = sg.to(stacked.device(), at::kFloat).contiguous();
+  TORCH_CHECK(hi.numel() == P && sf.numel() == P);
+  Tensor out = at::zeros({m, S}, stacked.options().dtype(at::kFloat));
+  dim3 grid(grid_for(P, BLOCK, 1024), m);
+  size_t lds = (size_t)S * sizeof(float);
+  DISPATCH_FT(stacked, {
+    count_sketch_kernel<elem_t><<<grid, BLOCK, lds, cur_stream()>>>(
+        (const elem_t*)stacked.data_ptr(), hi.data_ptr<int>(), sf.data_ptr<float>(), P,
+        (int)S, out.data_ptr<float>());
+  });
+  return out;
+}
+
+void sgd_step(Tensor p, Tensor g, double lr) {
+  check_flat(p, "p");
+  check_flat(g, "g");
+  TORCH_CHECK(p.numel() == g.numel() && p.scalar_type() == g.scalar_type());
+  int64_t P = p.numel();
+  int blocks = grid_for(P / 4, BLOCK);
+  DISPATCH_FT(p, {
+    sgd_step_kernel<elem_t><<<blocks, BLOCK, 0, cur_stream()>>>(
+        (elem_t*)p.data_ptr(), (const elem_t*)g.data_ptr(), (float)lr, P);
+  });
+}
+
+Tensor gaussian_inject(Tensor x, double sigma, int64_t seed, int64_t offset) {
+  check_flat(x, "x");
+  int64_t P = x.numel();
+  Tensor out = at::empty_like(x);
+  int blocks = grid_for((P + 3) / 4, BLOCK);
+  DISPATCH_FT(x, {
+    gaussian_inject_kernel<elem_t><<<blocks, BLOCK, 0, cur_stream()>>>(
+        (const elem_t*)x.data_ptr(), (elem_t*)out.data_ptr(), (float)sigma,
+        (unsigned long long)seed, (unsigned long long)offset, P);
+  });
+  return out;
+}
+
+Tensor scale_inject(Tensor x, double lam) {
+  check_flat(x, "x");
+  int64_t P = x.numel();
+  Tensor out = at::empty_like(x);
+  int blocks = grid_for(P / 4, BLOCK);
+  DISPATCH_FT(x, {
+    scale_kernel<elem_t><<<blocks, BLOCK, 0, cur_stream()>>>(
+        (const elem_t*)x.data_ptr(), (elem_t*)out.data_ptr(), (float)lam, P);
+  });
+  return out;
+}
+
+Tensor ce_loss_acc(Tensor logits, Tensor targets) {
+  check_flat(logits, "logits");
+  TORCH_CHECK(logits.dim() == 2);
+  Tensor tg = targets.to(logits.device(), at::kLong).contiguous();
+  int B = (int)logits.size(0), C = (int)logits.size(1);
+  Tensor out = at::zeros({2}, logits.options().dtype(at::kFloat));
+  int blocks = grid_for((int64_t)B, 4, 2048);
+  DISPATCH_FT(logits, {
+    ce_loss_acc_kernel<elem_t><<<blocks, BLOCK, 0, cur_stream()>>>(
+        (const elem_t*)logits.data_ptr(), tg.data_ptr<int64_t>(), B, C,
+        out.data_ptr<float>());
+  });
+  return out;
+}
+
+Tensor evidential_stats(Tensor logits, Tensor targets) {
+  check_flat(logits, "logits");
+  TORCH_CHECK(logits.dim() == 2);
+  Tensor tg = targets.to(logits.device(), at::kLong).contiguous();
+  int B = (int)logits.size(0), C = (int)logits.size(1);
+  Tensor out = at::zeros({4}, logits.options().dtype(at::kFloat));
+  int blocks = grid_for((int64_t)B, 4, 2048);
+  DISPATCH_FT(logits, {
+    evidential_stats_kernel<elem_t><<<blocks, BLOCK, 0, cur_stream()>>>(
+        (const elem_t*)logits.data_ptr(), tg.data_ptr<int64_t>(), B, C,
+        out.data_ptr<float>());
+  });
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("weighted_sum", &weighted_sum, "K1: out = sum_i w_i * x_i",
+        py::arg("stacked"), py::arg("w"), py::arg("out") = py::none());
+  m.def("pairwise_sq_dists", &pairwise_sq_dists, "K2: [m,m] squared L2 matrix");
+  m.def("row_norms", &row_norms, "K12: per-row L2 norms");
+  m.def("l2_dists_to", &l2_dists_to, "K2 variant: dists of rows to own");
+  m.def("count_sketch", &count_sketch, "K4: count-sketch projection");
+  m.def("sgd_step", &sgd_step, "K6: fused p -= lr*g");
+  m.def("gaussian_inject", &gaussian_inject, "K10: x + N(0, sigma^2) (Philox)");
+  m.def("scale_inject", &scale_inject, "K11: lam * x");
+  m.def("ce_loss_acc", &ce_loss_acc, "K7: [loss_sum, correct]");
+  m.def("evidential_stats", &evidential_stats,
+        "K8: [vacuity_sum, entropy_sum, strength_sum, correct]");
+}

@@ -1,0 +1,94 @@
+"""Neighbor-state exchange over torch.distributed (RCCL on ROCm, gloo on CPU).
+
+Replaces the reference's ZMQ PUSH/PULL + torch.save serialization
+(reference: murmura/distributed/messaging.py, node_process.py:227-276): states
+live in flat device buffers, RCCL moves raw bf16/fp32 bytes GPU-to-GPU over
+xGMI with zero serialization.
+
+Design:
+- Every GPU pair on an 8xMI355X box has a direct xGMI link, so per-edge
+  paired send/recv inside one ``batch_isend_irecv`` group is the right
+  primitive for sparse topologies (ring / k-regular / erdos / mobility G^t).
+- The per-round edge set is DATA, not communicator structure: dynamic
+  topologies need no re-initialization (the property the reference built its
+  whole ZMQ design around, paper.tex:552-556 — preserved here for free).
+- Fully-connected FedAvg folds the mean into one ``all_reduce`` of the flat
+  buffer (K1 fused into the collective).
+- Asymmetric per-round peer sets (DMTT's C_i^t) are made symmetric by an
+  all-gather of the N x N want-matrix before the grouped P2P (§7 hard-part 2).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence
+
+import torch
+import torch.distributed as dist
+from torch import Tensor
+
+from murmura_amd.topology.base import Topology
+
+
+def exchange_with_neighbors(
+    own: Tensor, neighbor_ids: Sequence[int], tag_base: int = 0
+) -> Dict[int, Tensor]:
+    """Symmetric exchange of ``own`` with each neighbor; returns
+    {neighbor_id: received flat state}. Every rank must call this with a
+    consistent (symmetric) edge set. Deadlock-free: all sends/recvs are posted
+    in one batch_isend_irecv group."""
+    if not neighbor_ids:
+        return {}
+    bufs = {j: torch.empty_like(own) for j in neighbor_ids}
+    ops: List[dist.P2POp] = []
+    # deterministic global order: send before recv per peer, peers sorted
+    for j in sorted(neighbor_ids):
+        ops.append(dist.P2POp(dist.isend, own, j))
+        ops.append(dist.P2POp(dist.irecv, bufs[j], j))
+    reqs = dist.batch_isend_irecv(ops)
+    for r in reqs:
+        r.wait()
+    return bufs
+
+
+def allreduce_mean(own: Tensor) -> Tensor:
+    """Fully-connected FedAvg fast path: global mean of flat states in one
+    RCCL all-reduce (ring over xGMI)."""
+    out = own.clone()
+    dist.all_reduce(out, op=dist.ReduceOp.SUM)
+    out.div_(dist.get_world_size())
+    return out
+
+
+def symmetrize_wants(want: Sequence[int], world_size: int, device) -> List[List[int]]:
+    """All-gather each rank's desired peer list and return the SYMMETRIC
+    exchange sets: i exchanges with j iff i wants j OR j wants i.
+
+    DMTT's collaborator sets are asymmetric; RCCL P2P requires both ends to
+    post matching ops, so the union set is agreed via one tiny all-gather of
+    an N-bit mask (8 bytes on an 8-GPU box)."""
+    mask = torch.zeros(world_size, dtype=torch.uint8, device=device)
+    for j in want:
+        mask[j] = 1
+    all_masks = [torch.zeros_like(mask) for _ in range(world_size)]
+    dist.all_gather(all_masks, mask)
+    m = torch.stack(all_masks).cpu()  # [N, N]; m[i][j] = i wants j
+    sym = m | m.t()
+    out: List[List[int]] = []
+    for i in range(world_size):
+        out.append([j for j in range(world_size) if j != i and sym[i][j]])
+    return out
+
+
+def gather_metrics(metrics: Optional[dict], dst: int = 0) -> Optional[List[dict]]:
+    """Host-side metrics gather (the monitor path — tiny dicts, not
+    perf-critical; reference: node_process.py:284-286)."""
+    world = dist.get_world_size()
+    gathered: Optional[List[Optional[dict]]] = (
+        [None] * world if dist.get_rank() == dst else None
+    )
+    dist.gather_object(metrics, gathered, dst=dst)
+    return gathered  # type: ignore[return-value]
+
+
+def make_topology_neighbors(topo: Topology, rank: int) -> List[int]:
+    return list(topo.neighbors[rank])

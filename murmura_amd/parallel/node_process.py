@@ -1,0 +1,241 @@
+"""Per-rank FL node loop over torch.distributed — one process per MI355X GPU.
+
+Replaces the reference's ZMQ NodeProcess (murmura/distributed/node_process.py:60-364):
+- rank == node_id; device = cuda:LOCAL_RANK (RCCL over xGMI) or cpu (gloo, tests)
+- rounds are synchronized by the collectives themselves (deterministic
+  intra-box), not by wall-clock windows; an optional round_duration_s budget is
+  kept only for parity experiments
+- per-node seeding seed + node_id (reference: node_process.py:113)
+- the per-round topology (static or mobility G^t) is computed locally and
+  identically on every rank from the shared seed — no coordination messages
+
+Round structure (mirrors Network.train semantics exactly so histories match
+the simulation backend on the same seeds):
+  1. local train (honest nodes)
+  2. snapshot flat state; compromised ranks attack their own snapshot
+     (they broadcast AND aggregate the attacked state, reference:
+     network.py:110-119)
+  3. exchange: all-reduce fast path (fedavg + fully-connected) or grouped
+     RCCL P2P along this round's edges
+  4. aggregate (HIP kernels) and apply
+  5. evaluate + gather metrics to rank 0 (host-side, tiny)
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+from torch.utils.data import DataLoader
+
+from murmura_amd.config.schema import Config
+from murmura_amd.core.network import HISTORY_KEYS, new_history
+from murmura_amd.core.node import Node
+from murmura_amd.parallel import exchange
+from murmura_amd.topology.dynamic import MobilityModel
+from murmura_amd.topology.generators import create_topology
+from murmura_amd.utils import factories
+from murmura_amd.utils.seed import set_seed
+
+
+def _resolve_backend(config: Config) -> str:
+    b = config.distributed.comm_backend
+    if b != "auto":
+        return b
+    return "nccl" if torch.cuda.is_available() else "gloo"
+
+
+def init_distributed(config: Config, rank: int, world_size: int) -> torch.device:
+    """Initialize the process group; returns this rank's device."""
+    backend = _resolve_backend(config)
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", config.distributed.master_addr)
+        os.environ.setdefault("MASTER_PORT", str(config.distributed.master_port))
+        dist.init_process_group(
+            backend=backend,
+            rank=rank,
+            world_size=world_size,
+            timeout=datetime.timedelta(seconds=300),
+        )
+    if backend == "nccl":
+        local = int(os.environ.get("LOCAL_RANK", rank % max(1, torch.cuda.device_count())))
+        device = torch.device(f"cuda:{local}")
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+    return device
+
+
+def build_node(config: Config, rank: int, device: torch.device) -> Node:
+    """Construct this rank's Node — mirrors Network.from_config per-node setup
+    (reference re-loads data per process too, node_process.py:333-364)."""
+    set_seed(config.experiment.seed + rank)
+    model_factory = factories.build_model_factory(config)
+    adapter = factories.build_dataset_adapter(config)
+    agg_factory = factories.build_aggregator_factory(config, model_factory)
+    criterion_factory = factories.build_criterion_factory(config)
+    evidential = factories.is_evidential(config)
+    dtype = factories.compute_dtype(config)
+
+    client_data = adapter.get_client_data(rank)
+    n_samples = len(client_data)
+    bs = min(config.training.batch_size, max(2, n_samples))
+    g = torch.Generator().manual_seed(config.experiment.seed + rank)
+    train_loader = DataLoader(
+        client_data, batch_size=bs, shuffle=True, drop_last=n_samples > bs, generator=g
+    )
+    test_loader = DataLoader(client_data, batch_size=bs, shuffle=False)
+    # per-node deterministic init: identical weights as the simulation backend
+    with torch.random.fork_rng(devices=[]):
+        torch.manual_seed(config.experiment.seed + rank)
+        model = model_factory()
+    return Node(
+        node_id=rank,
+        model=model,
+        train_loader=train_loader,
+        test_loader=test_loader,
+        aggregator=agg_factory(rank),
+        device=device,
+        criterion=criterion_factory() if criterion_factory else None,
+        evidential=evidential,
+        dtype=dtype,
+        model_factory=model_factory,
+    )
+
+
+class FLRoundLoop:
+    """The distributed round loop, factored so bench.py can time single rounds."""
+
+    def __init__(self, config: Config, rank: int, world_size: int, device: torch.device):
+        if config.topology.num_nodes != world_size:
+            raise ValueError(
+                f"world_size {world_size} != topology.num_nodes {config.topology.num_nodes}"
+            )
+        self.config = config
+        self.rank = rank
+        self.world = world_size
+        self.device = device
+        self.node = build_node(config, rank, device)
+        self.topology = create_topology(
+            config.topology.type,
+            world_size,
+            p=config.topology.p,
+            k=config.topology.k,
+            seed=config.topology.seed,
+        )
+        self.mobility = factories.build_mobility_model(config)
+        self.attack = factories.build_attack(config)
+        self.sketch_mode = config.distributed.sketch_wire_mode
+        self._fully_connected = all(
+            len(self.topology.neighbors[i]) == world_size - 1 for i in range(world_size)
+        )
+
+    def topology_at(self, round_num: int):
+        if self.mobility is not None:
+            return self.mobility.topology_at(round_num)
+        return self.topology
+
+    def _is_compromised(self, node_id: int) -> bool:
+        return self.attack is not None and self.attack.is_compromised(node_id)
+
+    def run_round(self, round_num: int) -> None:
+        cfg = self.config
+        topo = self.topology_at(round_num)
+        # 1. local training (honest nodes only; compromised stay frozen)
+        if not self._is_compromised(self.rank):
+            self.node.local_train(
+                epochs=cfg.training.local_epochs, lr=cfg.training.lr, round_num=round_num
+            )
+        # 2. snapshot + self-attack
+        own = self.node.get_state()
+        if self._is_compromised(self.rank):
+            own = self.attack.apply_attack(self.rank, own, round_num)
+
+        nbr_ids = list(topo.neighbors[self.rank])
+        use_allreduce = (
+            cfg.aggregation.algorithm == "fedavg"
+            and self.mobility is None
+            and self._fully_connected
+            and self.attack is None
+        )
+        if use_allreduce:
+            # K1 folded into the collective: new state = global mean
+            self.node.set_state(exchange.allreduce_mean(own))
+            return
+
+        # 3. grouped P2P along this round's edges
+        received = exchange.exchange_with_neighbors(own, nbr_ids)
+        if nbr_ids:
+            stacked = torch.stack([received[j] for j in nbr_ids], dim=0)
+        else:
+            stacked = own.new_zeros((0, own.numel()))
+        # 4. aggregate + apply
+        new_state = self.node.aggregate_with_neighbors(
+            own, stacked, neighbor_ids=nbr_ids, round_num=round_num
+        )
+        self.node.set_state(new_state)
+
+    def evaluate_round(self, round_num: int) -> Dict[str, float]:
+        res = self.node.evaluate()
+        out = {k: float(v) for k, v in res.items() if isinstance(v, torch.Tensor)}
+        out["round"] = round_num
+        out["node_id"] = self.rank
+        out["compromised"] = self._is_compromised(self.rank)
+        return out
+
+
+def _append_history(history: Dict[str, List[float]], round_num: int, rows: List[dict]) -> None:
+    import math
+
+    accs = [r["accuracy"] for r in rows]
+    losses = [r["loss"] for r in rows]
+    mean = sum(accs) / len(accs)
+    std = math.sqrt(sum((a - mean) ** 2 for a in accs) / len(accs))
+    honest = [r["accuracy"] for r in rows if not r["compromised"]]
+    comp = [r["accuracy"] for r in rows if r["compromised"]]
+    history["round"].append(round_num)
+    history["mean_accuracy"].append(mean)
+    history["std_accuracy"].append(std)
+    history["mean_loss"].append(sum(losses) / len(losses))
+    history["honest_accuracy"].append(sum(honest) / len(honest) if honest else 0.0)
+    history["compromised_accuracy"].append(sum(comp) / len(comp) if comp else 0.0)
+    for key, hkey in [
+        ("vacuity", "mean_vacuity"),
+        ("entropy", "mean_entropy"),
+        ("strength", "mean_strength"),
+    ]:
+        vals = [r[key] for r in rows if key in r]
+        history[hkey].append(sum(vals) / len(vals) if vals else 0.0)
+
+
+def run_node_process(
+    config: Config, rank: int, world_size: int, destroy_group: bool = True
+) -> Optional[Dict[str, List[float]]]:
+    """Full distributed run for one rank; rank 0 returns the history."""
+    device = init_distributed(config, rank, world_size)
+    if config.dmtt is not None:
+        from murmura_amd.dmtt.node_process import DMTTRoundLoop
+
+        loop: FLRoundLoop = DMTTRoundLoop(config, rank, world_size, device)
+    else:
+        loop = FLRoundLoop(config, rank, world_size, device)
+    history = new_history() if rank == 0 else None
+    for r in range(config.experiment.rounds):
+        loop.run_round(r)
+        metrics = loop.evaluate_round(r)
+        rows = exchange.gather_metrics(metrics, dst=0)
+        if rank == 0:
+            _append_history(history, r, rows)
+            if config.experiment.verbose:
+                print(
+                    f"[round {r}] acc={history['mean_accuracy'][-1]:.4f} "
+                    f"honest={history['honest_accuracy'][-1]:.4f}",
+                    flush=True,
+                )
+    dist.barrier()
+    if destroy_group:
+        dist.destroy_process_group()
+    return history
