@@ -1,0 +1,182 @@
+"""GPU numerics tests: every HIP kernel vs the plain-torch fp32 reference
+(murmura_amd/ops/reference.py) on the same inputs."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from murmura_amd import ops
+from murmura_amd.ops import reference as ref
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _require_native():
+    assert ops.native_available(), "HIP extension must be present on the GPU box"
+
+
+def _rand(shape, dtype=torch.float32, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randn(*shape, generator=g).to("cuda", dtype)
+
+
+P_ODD = 1_000_003  # odd size exercises the scalar tail
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("m,p", [(2, 1024), (8, P_ODD), (16, 65537)])
+def test_weighted_sum(dtype, m, p):
+    x = _rand((m, p), dtype)
+    w = torch.rand(m, device="cuda")
+    out = ops.weighted_sum(x, w)
+    expect = ref.weighted_sum(x.float().cpu(), w.cpu())
+    tol = 1e-4 if dtype == torch.float32 else 0.05
+    assert out.dtype == dtype
+    assert torch.allclose(out.float().cpu(), expect, atol=tol, rtol=tol)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("m", [2, 5, 8, 16])
+def test_pairwise_sq_dists(dtype, m):
+    x = _rand((m, P_ODD), dtype, seed=m)
+    d2 = ops.pairwise_sq_dists(x)
+    expect = ref.pairwise_sq_dists(x.float().cpu())
+    rel = (d2.cpu() - expect).abs().max() / expect.max().clamp_min(1e-6)
+    assert rel.item() < (1e-4 if dtype == torch.float32 else 2e-2)
+    assert torch.all(d2.diagonal().abs() / expect.max().clamp_min(1e-6) < 1e-4)
+
+
+def test_pairwise_large_m_fallback():
+    x = _rand((20, 10_001))
+    d2 = ops.pairwise_sq_dists(x)
+    expect = ref.pairwise_sq_dists(x.float().cpu())
+    assert torch.allclose(d2.cpu(), expect, rtol=1e-3, atol=1e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_row_norms_and_dists(dtype):
+    x = _rand((6, P_ODD), dtype, seed=3)
+    own = _rand((P_ODD,), dtype, seed=4)
+    n = ops.row_norms(x)
+    expect_n = ref.row_norms(x.float().cpu())
+    assert torch.allclose(n.cpu(), expect_n, rtol=1e-3)
+    d = ops.l2_dists_to(own, x)
+    expect_d = ref.l2_dists_to(own.float().cpu(), x.float().cpu())
+    assert torch.allclose(d.cpu(), expect_d, rtol=1e-3)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("sketch_size", [1000, 8192])
+def test_count_sketch(dtype, sketch_size):
+    m, p = 3, 500_009
+    x = _rand((m, p), dtype, seed=5)
+    h, s = ref.make_sketch_tables(p, sketch_size, seed=7, device=torch.device("cuda"))
+    out = ops.count_sketch(x, h, s, sketch_size)
+    expect = ref.count_sketch(x.float().cpu(), h.cpu(), s.cpu(), sketch_size)
+    scale = expect.abs().max().clamp_min(1e-6)
+    assert ((out.cpu() - expect).abs().max() / scale).item() < (
+        1e-4 if dtype == torch.float32 else 2e-2
+    )
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_sgd_step(dtype):
+    p = _rand((P_ODD,), dtype, seed=8)
+    g = _rand((P_ODD,), dtype, seed=9)
+    expect = (p.float() - 0.05 * g.float()).to(dtype)
+    ops.sgd_step(p, g, 0.05)
+    assert torch.allclose(p.float(), expect.float(), atol=1e-6)
+
+
+def test_gaussian_inject_statistics_and_determinism():
+    x = torch.zeros(2_000_003, device="cuda")
+    a = ops.gaussian_inject(x, 3.0, seed=11, offset=5)
+    b = ops.gaussian_inject(x, 3.0, seed=11, offset=5)
+    c = ops.gaussian_inject(x, 3.0, seed=11, offset=6)
+    assert torch.equal(a, b)
+    assert not torch.equal(a, c)
+    assert abs(a.mean().item()) < 0.01
+    assert abs(a.std().item() - 3.0) < 0.01
+    # normality sanity: ~68.3% within 1 sigma
+    frac = (a.abs() < 3.0).float().mean().item()
+    assert abs(frac - 0.683) < 0.01
+
+
+def test_gaussian_inject_bf16():
+    x = torch.zeros(1_000_000, device="cuda", dtype=torch.bfloat16)
+    a = ops.gaussian_inject(x, 2.0, seed=1, offset=0)
+    assert a.dtype == torch.bfloat16
+    assert abs(a.float().std().item() - 2.0) < 0.05
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_scale_inject(dtype):
+    x = _rand((P_ODD,), dtype)
+    out = ops.scale_inject(x, -5.0)
+    assert torch.allclose(out.float(), -5.0 * x.float(), rtol=1e-2, atol=1e-3)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("b,c", [(64, 10), (333, 62), (128, 7)])
+def test_ce_loss_acc(dtype, b, c):
+    logits = _rand((b, c), dtype, seed=b)
+    targets = torch.randint(0, c, (b,), device="cuda")
+    loss, correct = ops.ce_loss_acc(logits, targets)
+    eloss, ecorrect = ref.ce_loss_acc(logits.float().cpu(), targets.cpu())
+    assert correct.item() == ecorrect.item()
+    assert abs(loss.item() - eloss.item()) / max(1.0, eloss.item()) < (
+        1e-4 if dtype == torch.float32 else 1e-2
+    )
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("b,c", [(64, 6), (100, 12), (222, 100)])
+def test_evidential_stats(dtype, b, c):
+    logits = _rand((b, c), dtype, seed=b + c)
+    targets = torch.randint(0, c, (b,), device="cuda")
+    v, e, s, corr = ops.evidential_stats(logits, targets)
+    ev, ee, es, ec = ref.evidential_stats(logits.float().cpu(), targets.cpu())
+    tol = 1e-3 if dtype == torch.float32 else 2e-2
+    assert abs(v.item() - ev.item()) / max(1.0, ev.item()) < tol
+    assert abs(e.item() - ee.item()) / max(1.0, ee.item()) < tol
+    assert abs(s.item() - es.item()) / max(1.0, es.item()) < tol
+    assert corr.item() == ec.item()
+
+
+def test_gpu_aggregators_end_to_end():
+    """All six aggregators on GPU flat states produce finite outputs matching
+    the CPU torch path."""
+    import os
+
+    from murmura_amd.aggregation import (
+        BALANCEAggregator,
+        FedAvgAggregator,
+        KrumAggregator,
+        SketchguardAggregator,
+    )
+
+    P = 100_003
+    own = _rand((P,))
+    nbrs = _rand((5, P), seed=2)
+    for agg_cls, kw in [
+        (FedAvgAggregator, {}),
+        (KrumAggregator, {"num_compromised": 1}),
+        (BALANCEAggregator, {}),
+        (SketchguardAggregator, {"model_dim": P}),
+    ]:
+        out_gpu = agg_cls(**kw).aggregate(0, own, nbrs, round_num=1)
+        out_cpu = agg_cls(**kw).aggregate(0, own.cpu(), nbrs.cpu(), round_num=1)
+        assert torch.isfinite(out_gpu).all()
+        assert torch.allclose(out_gpu.cpu(), out_cpu, atol=1e-3, rtol=1e-3), agg_cls.__name__
+
+
+def test_native_required_on_gpu(monkeypatch):
+    """A cuda tensor with the extension 'missing' must raise, never silently
+    fall back to eager."""
+    import murmura_amd.ops as O
+
+    monkeypatch.setattr(O, "_EXT", None)
+    monkeypatch.setattr(O, "_EXT_ERR", "simulated-missing")
+    x = torch.randn(10, device="cuda")
+    with pytest.raises(RuntimeError, match="HIP extension"):
+        O.weighted_sum(x.view(1, -1).contiguous(), torch.ones(1, device="cuda"))
