@@ -52,13 +52,61 @@ class Node:
         self.evidential = evidential
         self.model_factory = model_factory
         self._eval_context: Optional[EvalContext] = None
+        # hipGraph round execution (device-resident shard + captured epoch/eval)
+        self._shard = None
+        self._train_graph = None
+        self._eval_graph = None
+
+    # ------------------------------------------------------------ hipGraphs
+    def _graphs_enabled(self) -> bool:
+        import os
+
+        return (
+            self.device.type == "cuda"
+            and os.environ.get("MURMURA_NO_GRAPHS") != "1"
+        )
+
+    def _get_shard(self, split: str = "train"):
+        from murmura_amd.core.gpu_round import DeviceShard
+
+        if self._shard is None:
+            self._shard = {}
+        if split not in self._shard:
+            if split == "test" and self.test_loader.dataset is self.train_loader.dataset:
+                # reference semantics: test = train data re-served (network.py:289-294)
+                self._shard[split] = self._get_shard("train")
+            else:
+                loader = self.train_loader if split == "train" else self.test_loader
+                self._shard[split] = DeviceShard.from_loader(loader, self.device, self.dtype)
+        return self._shard[split]
 
     # ------------------------------------------------------------ training
     def local_train(
         self, epochs: int = 1, lr: float = 0.01, round_num: int = 0
     ) -> Dict[str, float]:
         """Local SGD epochs. Skips batches with < 2 samples (BatchNorm needs
-        batch > 1 in training mode — reference: node.py:81)."""
+        batch > 1 in training mode — reference: node.py:81).
+
+        On GPU with a plain CE criterion the whole epoch runs as ONE hipGraph
+        replay over the device-resident shard (core/gpu_round.py); the eager
+        path below is the CPU oracle and the evidential/odd-shard fallback.
+        """
+        if (
+            self._graphs_enabled()
+            and not isinstance(self.criterion, EvidentialLoss)
+        ):
+            shard = self._get_shard("train")
+            bs = getattr(self.train_loader, "batch_size", None) or 32
+            if shard.n >= max(2, bs):
+                if self._train_graph is None:
+                    from murmura_amd.core.gpu_round import TrainGraph
+
+                    self._train_graph = TrainGraph(self, shard, bs)
+                total = torch.zeros((), device=self.device)
+                for _ in range(max(1, epochs)):
+                    total = total + self._train_graph.run_epoch(lr)
+                nb = self._train_graph.num_batches * max(1, epochs)
+                return {"loss": float(total.item() / nb), "num_batches": nb}
         self.model.train()
         gflat = self.store.ensure_grads()
         pflat = self.store.flat[: self.store.spec.param_numel]
@@ -85,6 +133,14 @@ class Node:
 
     # ------------------------------------------------------------ evaluation
     def evaluate(self) -> Dict[str, Tensor]:
+        if self._graphs_enabled():
+            shard = self._get_shard("test")
+            if self._eval_graph is None:
+                from murmura_amd.core.gpu_round import EvalGraph
+
+                bs = getattr(self.test_loader, "batch_size", None) or 64
+                self._eval_graph = EvalGraph(self, shard, bs, self.evidential)
+            return self._eval_graph.run()
         if self.evidential:
             return evaluate_evidential(self.model, self.test_loader, self.device, self.dtype)
         return evaluate_model(self.model, self.test_loader, self.device, self.dtype)

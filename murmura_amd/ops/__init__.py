@@ -117,6 +117,14 @@ def sgd_step(flat_params: Tensor, grad: Tensor, lr: float) -> None:
     ref.sgd_step(flat_params, grad, lr)
 
 
+def sgd_step_lrt(flat_params: Tensor, grad: Tensor, lr: Tensor) -> None:
+    """K6 with a device-tensor lr (hipGraph-capturable)."""
+    if _use_native(flat_params):
+        _EXT.sgd_step_lrt(flat_params, grad, lr)
+        return
+    flat_params.add_(grad * (-lr))
+
+
 # ------------------------------------------------------------------ K7
 def ce_loss_acc(logits: Tensor, targets: Tensor) -> Tuple[Tensor, Tensor]:
     if _use_native(logits):

@@ -271,6 +271,27 @@ __global__ void sgd_step_kernel(T* __restrict__ p, const T* __restrict__ g, floa
   }
 }
 
+// K6 variant: lr read from device memory (hipGraph-capturable with a
+// per-round lr update outside the graph)
+template <typename T>
+__global__ void sgd_step_lrt_kernel(T* __restrict__ p, const T* __restrict__ g,
+                                    const float* __restrict__ lr, int64_t P) {
+  const float l = *lr;
+  constexpr int N = Pack16<T>::N;
+  const int64_t nvec = P / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    Pack16<T> pv = reinterpret_cast<Pack16<T>*>(p)[v];
+    Pack16<T> gv = reinterpret_cast<const Pack16<T>*>(g)[v];
+#pragma unroll
+    for (int k = 0; k < N; ++k) from_f(pv.e[k], fmaf(-l, to_f(gv.e[k]), to_f(pv.e[k])));
+    reinterpret_cast<Pack16<T>*>(p)[v] = pv;
+  }
+  for (int64_t i = nvec * N + blockIdx.x * blockDim.x + threadIdx.x; i < P; i += stride) {
+    from_f(p[i], fmaf(-l, to_f(g[i]), to_f(p[i])));
+  }
+}
+
 // ================================================================== K10
 // Philox4x32-10 counter RNG + Box-Muller: out = x + N(0, sigma^2).
 __device__ __forceinline__ void philox_round(unsigned int& c0, unsigned int& c1,
@@ -618,6 +639,19 @@ void sgd_step(Tensor p, Tensor g, double lr) {
   });
 }
 
+void sgd_step_lrt(Tensor p, Tensor g, Tensor lr) {
+  check_flat(p, "p");
+  check_flat(g, "g");
+  TORCH_CHECK(p.numel() == g.numel() && p.scalar_type() == g.scalar_type());
+  TORCH_CHECK(lr.is_cuda() && lr.scalar_type() == at::kFloat && lr.numel() == 1);
+  int64_t P = p.numel();
+  int blocks = grid_for(P / 4, BLOCK);
+  DISPATCH_FT(p, {
+    sgd_step_lrt_kernel<elem_t><<<blocks, BLOCK, 0, cur_stream()>>>(
+        (elem_t*)p.data_ptr(), (const elem_t*)g.data_ptr(), lr.data_ptr<float>(), P);
+  });
+}
+
 Tensor gaussian_inject(Tensor x, double sigma, int64_t seed, int64_t offset) {
   check_flat(x, "x");
   int64_t P = x.numel();
@@ -683,6 +717,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2_dists_to", &l2_dists_to, "K2 variant: dists of rows to own");
   m.def("count_sketch", &count_sketch, "K4: count-sketch projection");
   m.def("sgd_step", &sgd_step, "K6: fused p -= lr*g");
+  m.def("sgd_step_lrt", &sgd_step_lrt, "K6: fused p -= lr*g, lr from device");
   m.def("gaussian_inject", &gaussian_inject, "K10: x + N(0, sigma^2) (Philox)");
   m.def("scale_inject", &scale_inject, "K11: lam * x");
   m.def("ce_loss_acc", &ce_loss_acc, "K7: [loss_sum, correct]");
