@@ -67,10 +67,13 @@ class TrainGraph:
         self.bs = batch_size
         self.num_batches = max(1, shard.n // batch_size)
         n_used = self.num_batches * batch_size
-        self.static_x = torch.empty(
+        # zero-init, NOT empty: warmup before the first shuffle must see valid
+        # class indices — garbage int64 targets make the NLL gather read out
+        # of bounds (an HSA memory fault, found the hard way on the GPU box)
+        self.static_x = torch.zeros(
             (n_used, *shard.x.shape[1:]), device=shard.device, dtype=shard.x.dtype
         )
-        self.static_y = torch.empty((n_used,), device=shard.device, dtype=shard.y.dtype)
+        self.static_y = torch.zeros((n_used,), device=shard.device, dtype=shard.y.dtype)
         self.loss_sum = torch.zeros((), device=shard.device, dtype=torch.float32)
         self.lr = torch.zeros((), device=shard.device, dtype=torch.float32)
         self.graph: Optional[torch.cuda.CUDAGraph] = None
@@ -118,10 +121,12 @@ class TrainGraph:
 
     def run_epoch(self, lr: float) -> Tensor:
         """Shuffle + replay one epoch; returns the summed loss (device scalar)."""
-        if self.graph is None:
-            self._capture()
+        # fill static buffers BEFORE the first capture: warmup executes real
+        # kernels and must see genuine inputs/targets
         self.shard.shuffled(self.static_x, self.static_y, self._gen)
         self.lr.fill_(lr)
+        if self.graph is None:
+            self._capture()
         self.graph.replay()
         return self.loss_sum
 
