@@ -53,7 +53,10 @@ class EvalContext:
         x, y = batch
         self.store.copy_from_flat(flat_state)
         self.store.model.eval()
-        logits = self.store.model(x.to(dtype=self.store.dtype))
+        x = x.to(dtype=self.store.dtype)
+        if self.store.channels_last and x.dim() == 4:
+            x = x.contiguous(memory_format=torch.channels_last)
+        logits = self.store.model(x)
         loss_sum, _ = ops.ce_loss_acc(logits, y)
         return loss_sum / max(1, x.shape[0])
 
@@ -70,6 +73,8 @@ class EvalContext:
             if seen >= max_samples:
                 break
             x = x.to(self.device, dtype=self.store.dtype)
+            if self.store.channels_last and x.dim() == 4:
+                x = x.contiguous(memory_format=torch.channels_last)
             y = y.to(self.device)
             if seen + x.shape[0] > max_samples:
                 take = max_samples - seen

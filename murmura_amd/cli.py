@@ -49,7 +49,8 @@ def _run_simulation(config, verbose: bool):
     from murmura_amd.core.network import Network
     from murmura_amd.utils import factories, seed
 
-    seed.set_seed(config.experiment.seed)
+    seed.set_seed(config.experiment.seed,
+                  deterministic_kernels=config.compute.deterministic_kernels)
     model_factory = factories.build_model_factory(config)
     adapter = factories.build_dataset_adapter(config)
     agg_factory = factories.build_aggregator_factory(config, model_factory)

@@ -127,6 +127,12 @@ class ComputeConfig(_Strict):
     device: str = "auto"
     # use fused HIP kernels when on GPU (fail loudly if extension missing)
     native_kernels: bool = True
+    # force MIOpen deterministic conv algorithms (8-17x slower on MI355X;
+    # seeded RNGs alone already reproduce training curves)
+    deterministic_kernels: bool = False
+    # NHWC weights/activations for conv models (removes MIOpen's internal
+    # batched_transpose kernels on CDNA4)
+    channels_last: bool = True
 
 
 class Config(_Strict):

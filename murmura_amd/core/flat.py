@@ -85,26 +85,58 @@ class FlatParamStore:
     ``ensure_grads``) gradients accumulate into the contiguous ``grad_flat``
     so one fused kernel performs the whole SGD step (K6)."""
 
-    def __init__(self, model: nn.Module, device: torch.device, dtype: torch.dtype):
+    def __init__(
+        self,
+        model: nn.Module,
+        device: torch.device,
+        dtype: torch.dtype,
+        channels_last: bool = False,
+    ):
         model = model.to(device=device, dtype=dtype)
         self.model = model
         self.device = torch.device(device)
         self.dtype = dtype
+        # NHWC storage for 4-D (conv) params: the flat slice holds the weight
+        # in (O, H, W, I) order and the bound param is the permuted view, so
+        # MIOpen sees channels_last weights (no internal batched_transpose on
+        # CDNA4) while aggregation/RCCL still see one contiguous P-vector.
+        # Layout is identical across nodes, so cross-node math is unaffected.
+        self.channels_last = channels_last
         self.spec = FlatParamSpec.from_model(model)
         self.flat = torch.empty(self.spec.total_numel, device=device, dtype=dtype)
         self.grad_flat: Optional[Tensor] = None
         self._bind(copy=True)
+
+    def _is_nhwc(self, e: FlatEntry) -> bool:
+        return self.channels_last and e.is_param and len(e.shape) == 4
+
+    def _bound_view(self, buf: Tensor, e: FlatEntry) -> Tensor:
+        """The NCHW-shaped (possibly channels_last-strided) tensor over the
+        flat slice for entry ``e``."""
+        sl = buf[e.offset : e.offset + e.numel]
+        if self._is_nhwc(e):
+            o, i, h, w = e.shape
+            return sl.view(o, h, w, i).permute(0, 3, 1, 2)
+        return sl.view(e.shape)
+
+    @torch.no_grad()
+    def _write_entry(self, buf: Tensor, e: FlatEntry, src: Tensor) -> None:
+        sl = buf[e.offset : e.offset + e.numel]
+        if self._is_nhwc(e):
+            o, i, h, w = e.shape
+            sl.view(o, h, w, i).copy_(src.detach().permute(0, 2, 3, 1))
+        else:
+            sl.view(e.shape).copy_(src.detach())
 
     def _bind(self, copy: bool) -> None:
         """Point every param/float-buffer at its view of ``self.flat``."""
         params = dict(self.model.named_parameters())
         buffers = dict(self.model.named_buffers())
         for e in self.spec:
-            view = self.flat[e.offset : e.offset + e.numel].view(e.shape)
             src = params[e.name] if e.is_param else buffers[e.name]
             if copy:
-                with torch.no_grad():
-                    view.copy_(src.detach())
+                self._write_entry(self.flat, e, src)
+            view = self._bound_view(self.flat, e)
             if e.is_param:
                 params[e.name].data = view
             else:
@@ -115,7 +147,9 @@ class FlatParamStore:
 
     def ensure_grads(self) -> Tensor:
         """Allocate the flat grad buffer (param prefix only) and point every
-        param's ``.grad`` at its slice so autograd accumulates in place."""
+        param's ``.grad`` at its slice so autograd accumulates in place.
+        Grad views carry the same (possibly channels_last) strides as their
+        params — autograd requires grad layout to match."""
         if self.grad_flat is None:
             self.grad_flat = torch.zeros(
                 self.spec.param_numel, device=self.device, dtype=self.dtype
@@ -123,9 +157,7 @@ class FlatParamStore:
         params = dict(self.model.named_parameters())
         for e in self.spec:
             if e.is_param:
-                params[e.name].grad = self.grad_flat[e.offset : e.offset + e.numel].view(
-                    e.shape
-                )
+                params[e.name].grad = self._bound_view(self.grad_flat, e)
         return self.grad_flat
 
     def zero_grad(self) -> None:
@@ -146,10 +178,11 @@ class FlatParamStore:
 
     def to_state_dict(self) -> Dict[str, Tensor]:
         """Full state dict (flat views reshaped + non-float buffers), for
-        checkpointing and reference-API parity."""
+        checkpointing and reference-API parity. Tensors are NCHW-shaped
+        (channels_last entries are permuted views over the flat slice)."""
         out: Dict[str, Tensor] = {}
         for e in self.spec:
-            out[e.name] = self.flat[e.offset : e.offset + e.numel].view(e.shape)
+            out[e.name] = self._bound_view(self.flat, e)
         for name, b in self.model.named_buffers():
             if not torch.is_floating_point(b):
                 out[name] = b
@@ -159,8 +192,8 @@ class FlatParamStore:
     def load_state_dict(self, state: Dict[str, Tensor]) -> None:
         for e in self.spec:
             if e.name in state:
-                self.flat[e.offset : e.offset + e.numel].copy_(
-                    state[e.name].reshape(-1).to(device=self.device, dtype=self.dtype)
+                self._write_entry(
+                    self.flat, e, state[e.name].to(device=self.device, dtype=self.dtype)
                 )
         for name, b in self.model.named_buffers():
             if not torch.is_floating_point(b) and name in state:

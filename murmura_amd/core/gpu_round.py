@@ -35,27 +35,32 @@ from murmura_amd import ops
 class DeviceShard:
     """A node's local dataset materialized on its GPU."""
 
-    def __init__(self, x: Tensor, y: Tensor, device: torch.device, dtype: torch.dtype):
+    def __init__(self, x: Tensor, y: Tensor, device: torch.device, dtype: torch.dtype,
+                 channels_last: bool = False):
         self.x = x.to(device=device, dtype=dtype, non_blocking=True).contiguous()
+        if channels_last and self.x.dim() == 4:
+            self.x = self.x.contiguous(memory_format=torch.channels_last)
         self.y = y.to(device=device, non_blocking=True).contiguous()
         self.device = device
         self.n = self.x.shape[0]
+        self.channels_last = channels_last and self.x.dim() == 4
 
     @classmethod
-    def from_loader(cls, loader, device: torch.device, dtype: torch.dtype) -> "DeviceShard":
+    def from_loader(cls, loader, device: torch.device, dtype: torch.dtype,
+                    channels_last: bool = False) -> "DeviceShard":
         xs, ys = [], []
         for x, y in loader:
             xs.append(x)
             ys.append(y)
-        return cls(torch.cat(xs), torch.cat(ys), device, dtype)
+        return cls(torch.cat(xs), torch.cat(ys), device, dtype, channels_last)
 
     def shuffled(self, out_x: Tensor, out_y: Tensor, generator: torch.Generator) -> None:
         """Gather a fresh permutation of the shard into the graph's static
         input buffers (one kernel each)."""
         n = out_x.shape[0]
         perm = torch.randperm(self.n, generator=generator)[:n].to(self.device)
-        torch.index_select(self.x, 0, perm, out=out_x)
-        torch.index_select(self.y, 0, perm, out=out_y)
+        out_x.copy_(self.x.index_select(0, perm))
+        out_y.copy_(self.y.index_select(0, perm))
 
 
 class TrainGraph:
@@ -73,6 +78,8 @@ class TrainGraph:
         self.static_x = torch.zeros(
             (n_used, *shard.x.shape[1:]), device=shard.device, dtype=shard.x.dtype
         )
+        if shard.channels_last:
+            self.static_x = self.static_x.contiguous(memory_format=torch.channels_last)
         self.static_y = torch.zeros((n_used,), device=shard.device, dtype=shard.y.dtype)
         self.loss_sum = torch.zeros((), device=shard.device, dtype=torch.float32)
         self.lr = torch.zeros((), device=shard.device, dtype=torch.float32)

@@ -202,6 +202,7 @@ class Network:
             )
         attack = build_attack(config)
         dtype = torch.bfloat16 if config.compute.dtype == "bf16" else torch.float32
+        channels_last = config.compute.channels_last
 
         def seeded_model(i: int) -> torch.nn.Module:
             # per-node deterministic init: identical weights in the simulation
@@ -237,6 +238,7 @@ class Network:
                     evidential=evidential,
                     dtype=dtype,
                     model_factory=model_factory,
+                    channels_last=channels_last,
                 )
             )
         return cls(nodes, topo, attack=attack, mobility=mobility)

@@ -39,11 +39,18 @@ class Node:
         evidential: bool = False,
         dtype: torch.dtype = torch.float32,
         model_factory: Optional[Callable[[], nn.Module]] = None,
+        channels_last: bool = True,
     ) -> None:
         self.node_id = node_id
         self.device = torch.device(device)
         self.dtype = dtype
-        self.store = FlatParamStore(model, self.device, dtype)
+        # NHWC for conv models on GPU (MIOpen's native layout on CDNA4)
+        self.channels_last = (
+            channels_last
+            and self.device.type == "cuda"
+            and any(p.dim() == 4 for p in model.parameters())
+        )
+        self.store = FlatParamStore(model, self.device, dtype, channels_last=self.channels_last)
         self.model = self.store.model
         self.train_loader = train_loader
         self.test_loader = test_loader
@@ -77,7 +84,9 @@ class Node:
                 self._shard[split] = self._get_shard("train")
             else:
                 loader = self.train_loader if split == "train" else self.test_loader
-                self._shard[split] = DeviceShard.from_loader(loader, self.device, self.dtype)
+                self._shard[split] = DeviceShard.from_loader(
+                    loader, self.device, self.dtype, channels_last=self.channels_last
+                )
         return self._shard[split]
 
     # ------------------------------------------------------------ training
@@ -117,6 +126,8 @@ class Node:
                 if x.shape[0] < 2:
                     continue
                 x = x.to(device=self.device, dtype=self.dtype, non_blocking=True)
+                if self.channels_last and x.dim() == 4:
+                    x = x.contiguous(memory_format=torch.channels_last)
                 y = y.to(self.device, non_blocking=True)
                 self.store.zero_grad()
                 out = self.model(x)
@@ -158,7 +169,10 @@ class Node:
         if self._eval_context is None:
             if self.model_factory is None:
                 return None
-            scratch = FlatParamStore(self.model_factory(), self.device, self.dtype)
+            scratch = FlatParamStore(
+                self.model_factory(), self.device, self.dtype,
+                channels_last=self.channels_last,
+            )
             self._eval_context = EvalContext(
                 scratch, self.train_loader, self.device, self.evidential
             )

@@ -72,7 +72,8 @@ def init_distributed(config: Config, rank: int, world_size: int) -> torch.device
 def build_node(config: Config, rank: int, device: torch.device) -> Node:
     """Construct this rank's Node — mirrors Network.from_config per-node setup
     (reference re-loads data per process too, node_process.py:333-364)."""
-    set_seed(config.experiment.seed + rank)
+    set_seed(config.experiment.seed + rank,
+             deterministic_kernels=config.compute.deterministic_kernels)
     model_factory = factories.build_model_factory(config)
     adapter = factories.build_dataset_adapter(config)
     agg_factory = factories.build_aggregator_factory(config, model_factory)
@@ -103,6 +104,7 @@ def build_node(config: Config, rank: int, device: torch.device) -> Node:
         evidential=evidential,
         dtype=dtype,
         model_factory=model_factory,
+        channels_last=config.compute.channels_last,
     )
 
 
