@@ -45,7 +45,10 @@ def parse_args():
                    help="default: fully (fedavg) / k-regular (krum)")
     p.add_argument("--attack", default="none", choices=["none", "gaussian", "directed"])
     p.add_argument("--model", default="resnet18",
-                   choices=["resnet18", "femnist-baseline", "femnist-xlarge", "mlp"])
+                   choices=["resnet18", "femnist-baseline", "femnist-xlarge", "mlp",
+                            "wide100m"])
+    p.add_argument("--sketch-wire", action="store_true",
+                   help="sketchguard: exchange 4KB sketches first, full states only with accepted neighbors")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--shard", type=int, default=2048, help="samples per node")
     p.add_argument("--batch-size", type=int, default=64)
@@ -67,6 +70,11 @@ def build_config(args, world):
                            {"image_shape": [1, 28, 28], "num_classes": 62}),
         "mlp": ("models.mlp", {"in_features": 32, "hidden": 64, "num_classes": 4},
                 {"num_features": 32, "num_classes": 4}),
+        # ~100M-param MLP: the Sketchguard bandwidth-bound xGMI showcase
+        # (BASELINE.json config 4)
+        "wide100m": ("models.widemlp",
+                     {"in_features": 4096, "hidden": 12288, "num_classes": 62},
+                     {"num_features": 4096, "num_classes": 62}),
     }
     factory, mparams, dparams = model_cfgs[args.model]
     topo = args.topology or ("k-regular" if args.algo == "krum" else "fully")
@@ -94,6 +102,7 @@ def build_config(args, world):
                             "partition": "iid", **dparams}},
         "model": {"factory": factory, "params": mparams},
         "backend": "rccl",
+        "distributed": {"sketch_wire_mode": bool(getattr(args, "sketch_wire", False))},
         "compute": {"dtype": args.dtype, "native_kernels": True},
     })
 
@@ -161,6 +170,9 @@ def main():
 
     ms_per_step = elapsed / args.steps * 1000.0
     rounds_per_sec = args.steps / elapsed
+
+    if os.environ.get("MURMURA_TIMING") == "1" and rank == 0:
+        print(f"phase timings (mean ms): {loop.timer.summary()}", file=sys.stderr)
 
     if rank == 0:
         result = {

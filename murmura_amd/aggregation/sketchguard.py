@@ -66,6 +66,22 @@ class SketchguardAggregator(Aggregator):
         h, s = self._get_tables(flat_state.device)
         return ops.count_sketch(flat_state, h, s, self.sketch_size)
 
+    def wire_filter(
+        self, own_sketch: Tensor, neighbor_sketches: Tensor, round_num: int
+    ) -> Tensor:
+        """The filtering decision from sketches alone (no full states) — the
+        basis of sketch-first wire exchange: returns the boolean accept mask
+        over neighbors. Uses the same threshold math as ``aggregate``."""
+        dists = (neighbor_sketches.float() - own_sketch.float().unsqueeze(0)).norm(dim=1)
+        t_frac = round_num / max(1, self.total_rounds)
+        threshold = (
+            self.gamma
+            * math.exp(-self.kappa * t_frac)
+            * self._attack_factor(own_sketch.device)
+            * own_sketch.float().norm()
+        )
+        return dists <= threshold
+
     def _attack_factor(self, device: torch.device) -> Tensor:
         """1.5 if the mean of the last 3 acceptance rates < 0.3 else 1.0,
         as a device scalar (no host sync)."""
@@ -86,8 +102,13 @@ class SketchguardAggregator(Aggregator):
         neighbor_states: Tensor,
         round_num: int = 0,
         neighbor_sketches: Optional[Tensor] = None,
+        full_neighbor_count: Optional[int] = None,
         **ctx: Any,
     ) -> Tensor:
+        """``full_neighbor_count``: in sketch-wire mode the caller prefilters
+        and passes only the accepted subset; the acceptance-rate history (which
+        drives the adaptive attack factor) must still be measured against the
+        FULL neighbor count for parity with the unfiltered path."""
         k = neighbor_states.shape[0]
         if k == 0:
             return own_state.clone()
@@ -106,7 +127,10 @@ class SketchguardAggregator(Aggregator):
         )
         accept = dists <= threshold
         w = accept_weights(accept, dists, self.min_neighbors)
-        self._acceptance_history.append(accept.float().mean())
+        if full_neighbor_count is not None and full_neighbor_count > 0:
+            self._acceptance_history.append(accept.sum().float() / full_neighbor_count)
+        else:
+            self._acceptance_history.append(accept.float().mean())
         return blend(own_state, neighbor_states, w, self.alpha)
 
     def get_statistics(self) -> Dict[str, Any]:

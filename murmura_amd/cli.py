@@ -25,6 +25,11 @@ def run(
     device: Optional[str] = typer.Option(None, "--device", help="device override"),
     verbose: bool = typer.Option(True, "--verbose/--quiet"),
     output: Optional[Path] = typer.Option(None, "--output", help="write history JSON here"),
+    checkpoint: Optional[Path] = typer.Option(
+        None, "--checkpoint", help="checkpoint file (written every --checkpoint-every rounds)"
+    ),
+    checkpoint_every: int = typer.Option(0, "--checkpoint-every"),
+    resume: bool = typer.Option(False, "--resume", help="resume from --checkpoint"),
 ) -> None:
     """Run an experiment from a config (simulation or distributed backend)."""
     from murmura_amd.config.loader import load_config
@@ -33,7 +38,8 @@ def run(
     if device is not None:
         config.compute.device = device
     if config.backend == "simulation":
-        history = _run_simulation(config, verbose)
+        history = _run_simulation(config, verbose, checkpoint=checkpoint,
+                                  checkpoint_every=checkpoint_every, resume=resume)
     else:
         from murmura_amd.parallel.runner import DistributedRunner
 
@@ -43,7 +49,8 @@ def run(
     _display_results(history, config)
 
 
-def _run_simulation(config, verbose: bool):
+def _run_simulation(config, verbose: bool, checkpoint=None, checkpoint_every=0,
+                    resume=False):
     import torch
 
     from murmura_amd.core.network import Network
@@ -75,11 +82,19 @@ def _run_simulation(config, verbose: bool):
             f"{config.topology.type} topology, {config.aggregation.algorithm}, "
             f"backend=simulation"
         )
+    start_round = 0
+    if resume and checkpoint is not None:
+        start_round = network.resume_from(str(checkpoint))
+        if verbose:
+            console.print(f"resumed from {checkpoint} at round {start_round}")
     return network.train(
         rounds=config.experiment.rounds,
         local_epochs=config.training.local_epochs,
         lr=config.training.lr,
         verbose=verbose and config.experiment.verbose,
+        checkpoint_path=str(checkpoint) if checkpoint else None,
+        checkpoint_every=checkpoint_every,
+        start_round=start_round,
     )
 
 

@@ -72,14 +72,29 @@ class Network:
         lr: float = 0.01,
         eval_every: int = 1,
         verbose: bool = False,
+        checkpoint_path: Optional[str] = None,
+        checkpoint_every: int = 0,
+        start_round: int = 0,
     ) -> Dict[str, List[float]]:
-        for r in range(rounds):
+        from murmura_amd.utils import checkpoint as ckpt
+
+        for r in range(start_round, rounds):
             topo = self._topology_at(r)
             self._local_training_step(r, local_epochs, lr)
             self._aggregation_step(r, topo)
             if eval_every and (r % eval_every == 0 or r == rounds - 1):
                 self._evaluation_step(r, verbose)
+            if checkpoint_path and checkpoint_every and (
+                (r + 1) % checkpoint_every == 0 or r == rounds - 1
+            ):
+                ckpt.save_checkpoint(checkpoint_path, r, self.nodes, self.history)
         return self.history
+
+    def resume_from(self, checkpoint_path: str) -> int:
+        """Restore node states + history; returns the next round index."""
+        from murmura_amd.utils import checkpoint as ckpt
+
+        return ckpt.restore_network(self, ckpt.load_checkpoint(checkpoint_path))
 
     def _topology_at(self, round_num: int) -> Topology:
         if self.mobility is not None:

@@ -145,6 +145,27 @@ class ResNet18(nn.Module):
         return self.fc(out)
 
 
+class WideMLP(nn.Module):
+    """~100M-param MLP (in 4096 -> 12288 -> 4096 -> classes): the
+    bandwidth-bound Sketchguard showcase model (BASELINE.json config 4) —
+    P ~ 100M floats means each neighbor state is 400 MB fp32 / 200 MB bf16
+    on the xGMI wire."""
+
+    def __init__(self, in_features: int = 4096, hidden: int = 12288,
+                 num_classes: int = 62):
+        super().__init__()
+        self.net = nn.Sequential(
+            nn.Linear(in_features, hidden),
+            nn.ReLU(),
+            nn.Linear(hidden, in_features),
+            nn.ReLU(),
+            nn.Linear(in_features, num_classes),
+        )
+
+    def forward(self, x: Tensor) -> Tensor:
+        return self.net(x)
+
+
 def count_params(model: nn.Module) -> int:
     return sum(p.numel() for p in model.parameters())
 
@@ -154,4 +175,5 @@ MODEL_FACTORIES: Dict[str, Callable[..., nn.Module]] = {
     "models.femnist": FEMNISTModel,
     "models.celeba": CelebAModel,
     "models.resnet18": ResNet18,
+    "models.widemlp": WideMLP,
 }

@@ -39,6 +39,7 @@ from murmura_amd.topology.dynamic import MobilityModel
 from murmura_amd.topology.generators import create_topology
 from murmura_amd.utils import factories
 from murmura_amd.utils.seed import set_seed
+from murmura_amd.utils.timing import PhaseTimer
 
 
 def _resolve_backend(config: Config) -> str:
@@ -131,6 +132,7 @@ class FLRoundLoop:
         self.mobility = factories.build_mobility_model(config)
         self.attack = factories.build_attack(config)
         self.sketch_mode = config.distributed.sketch_wire_mode
+        self.timer = PhaseTimer(device)
         self._fully_connected = all(
             len(self.topology.neighbors[i]) == world_size - 1 for i in range(world_size)
         )
@@ -147,14 +149,17 @@ class FLRoundLoop:
         cfg = self.config
         topo = self.topology_at(round_num)
         # 1. local training (honest nodes only; compromised stay frozen)
-        if not self._is_compromised(self.rank):
-            self.node.local_train(
-                epochs=cfg.training.local_epochs, lr=cfg.training.lr, round_num=round_num
-            )
+        with self.timer.phase("train"):
+            if not self._is_compromised(self.rank):
+                self.node.local_train(
+                    epochs=cfg.training.local_epochs, lr=cfg.training.lr,
+                    round_num=round_num,
+                )
         # 2. snapshot + self-attack
-        own = self.node.get_state()
-        if self._is_compromised(self.rank):
-            own = self.attack.apply_attack(self.rank, own, round_num)
+        with self.timer.phase("snapshot_attack"):
+            own = self.node.get_state()
+            if self._is_compromised(self.rank):
+                own = self.attack.apply_attack(self.rank, own, round_num)
 
         nbr_ids = list(topo.neighbors[self.rank])
         use_allreduce = (
@@ -165,23 +170,92 @@ class FLRoundLoop:
         )
         if use_allreduce:
             # K1 folded into the collective: new state = global mean
-            self.node.set_state(exchange.allreduce_mean(own))
+            with self.timer.phase("exchange"):
+                new_state = exchange.allreduce_mean(own)
+            self.node.set_state(new_state)
+            return
+
+        if self.sketch_mode and cfg.aggregation.algorithm == "sketchguard":
+            self._run_round_sketch_wire(own, nbr_ids, round_num)
             return
 
         # 3. grouped P2P along this round's edges
-        received = exchange.exchange_with_neighbors(own, nbr_ids)
+        with self.timer.phase("exchange"):
+            received = exchange.exchange_with_neighbors(own, nbr_ids)
         if nbr_ids:
             stacked = torch.stack([received[j] for j in nbr_ids], dim=0)
         else:
             stacked = own.new_zeros((0, own.numel()))
         # 4. aggregate + apply
+        with self.timer.phase("aggregate"):
+            new_state = self.node.aggregate_with_neighbors(
+                own, stacked, neighbor_ids=nbr_ids, round_num=round_num
+            )
+            self.node.set_state(new_state)
+
+    def _run_round_sketch_wire(self, own, nbr_ids, round_num: int) -> None:
+        """Sketchguard sketch-first exchange (the comm-saving mode the
+        reference left latent, sketchguard.py:114-132):
+
+        1. all-gather the 4 KB Count-Sketches (K4 kernel) — N x S floats
+        2. filter locally in sketch space (identical threshold math)
+        3. move FULL P-vectors over xGMI only along accepted edges
+           (plus the closest-neighbor fallback edge)
+
+        Rejected neighbors' states never cross the wire: per round per node
+        the P2P volume drops from |N| x P floats to |accepted| x P + N x S.
+        """
+        import torch as _t
+
+        agg = self.node.aggregator
+        own_sketch = agg.get_sketch(own)  # [S]
+        all_sk = [_t.empty_like(own_sketch) for _ in range(self.world)]
+        dist.all_gather(all_sk, own_sketch.contiguous())
+        all_sk = _t.stack(all_sk)  # [N, S]
+        if not nbr_ids:
+            self.node.set_state(own)
+            return
+        nbr_sk = all_sk[_t.tensor(nbr_ids, device=own.device)]
+        accept = agg.wire_filter(own_sketch, nbr_sk, round_num)
+        # fallback: always fetch the sketch-closest neighbor so the
+        # min_neighbors fallback (balance.py:133-135 semantics) has its state
+        dists = (nbr_sk.float() - own_sketch.float().unsqueeze(0)).norm(dim=1)
+        closest = int(dists.argmin().item())
+        accept_idx = [i for i, a in enumerate(accept.tolist()) if a]
+        if closest not in accept_idx:
+            accept_idx.append(closest)
+        want = [nbr_ids[i] for i in accept_idx]
+        self.last_wire_stats = {
+            "accepted": len(accept_idx),
+            "neighbors": len(nbr_ids),
+            "sketch_bytes": int(all_sk.numel() * 4),
+            "state_bytes_exchanged": int(len(want) * own.numel() * own.element_size()),
+            "state_bytes_saved": int(
+                (len(nbr_ids) - len(want)) * own.numel() * own.element_size()
+            ),
+        }
+        sym = exchange.symmetrize_wants(want, self.world, own.device)
+        received = exchange.exchange_with_neighbors(own, sym[self.rank])
+        use = [j for j in want if j in received]
+        if use:
+            stacked = _t.stack([received[j] for j in use], dim=0)
+            use_sk = all_sk[_t.tensor(use, device=own.device)]
+        else:
+            stacked = own.new_zeros((0, own.numel()))
+            use_sk = None
         new_state = self.node.aggregate_with_neighbors(
-            own, stacked, neighbor_ids=nbr_ids, round_num=round_num
+            own,
+            stacked,
+            neighbor_ids=use,
+            round_num=round_num,
+            neighbor_sketches=use_sk,
+            full_neighbor_count=len(nbr_ids),
         )
         self.node.set_state(new_state)
 
     def evaluate_round(self, round_num: int) -> Dict[str, float]:
-        res = self.node.evaluate()
+        with self.timer.phase("evaluate"):
+            res = self.node.evaluate()
         out = {k: float(v) for k, v in res.items() if isinstance(v, torch.Tensor)}
         out["round"] = round_num
         out["node_id"] = self.rank

@@ -118,3 +118,16 @@ def test_distributed_dmtt_topology_liar():
     })
     h = _run_distributed(cfg, 3, 29615)
     assert len(h["round"]) == 2
+
+
+def test_sketch_wire_mode_matches_full_exchange():
+    """Sketch-first wire exchange must produce the same history as full-state
+    exchange (the filter decision is identical, only the wire traffic drops)."""
+    base = _base_config(3, algo="sketchguard", topo="fully", rounds=3, attack=True)
+    h_full = _run_distributed(base, 3, 29620)
+    wired = json.loads(json.dumps(base))
+    wired["distributed"]["sketch_wire_mode"] = True
+    h_wire = _run_distributed(wired, 3, 29621)
+    for key in ["mean_accuracy", "mean_loss", "honest_accuracy"]:
+        for a, b in zip(h_wire[key], h_full[key]):
+            assert a == pytest.approx(b, abs=2e-3), (key, h_wire[key], h_full[key])
