@@ -109,6 +109,17 @@ def build_node(config: Config, rank: int, device: torch.device) -> Node:
     )
 
 
+def _drop_stragglers(topo, ready: List[bool]):
+    """Edges restricted to ready nodes (stragglers neither send nor receive
+    this round — everyone computes the same reduced edge set locally)."""
+    from murmura_amd.topology.base import Topology
+
+    edges = {
+        (i, j) for i, j in topo.edges if ready[i] and ready[j]
+    }
+    return Topology.from_edges(topo.num_nodes, edges)
+
+
 class FLRoundLoop:
     """The distributed round loop, factored so bench.py can time single rounds."""
 
@@ -145,16 +156,45 @@ class FLRoundLoop:
     def _is_compromised(self, node_id: int) -> bool:
         return self.attack is not None and self.attack.is_compromised(node_id)
 
+    def _ready_mask(self, train_seconds: float) -> Optional[List[bool]]:
+        """Straggler semantics (reference: node_process.py:210-217 skips the
+        exchange when training overruns the round budget; peers aggregate with
+        whatever arrived). With collectives the equivalent is agreed
+        deterministically: one tiny all-gather of a ready bit; overrun nodes
+        are dropped from EVERYONE's edge set this round. Disabled when
+        round_duration_s == 0 (intra-box deterministic-synchronous mode)."""
+        budget = self.config.distributed.round_duration_s
+        if budget <= 0:
+            return None
+        ready = torch.tensor(
+            [1 if train_seconds <= budget else 0], dtype=torch.uint8
+        )
+        if self.device.type == "cuda":
+            ready = ready.to(self.device)
+        out = [torch.zeros_like(ready) for _ in range(self.world)]
+        dist.all_gather(out, ready)
+        return [bool(t.item()) for t in out]
+
     def run_round(self, round_num: int) -> None:
+        import time as _time
+
         cfg = self.config
         topo = self.topology_at(round_num)
         # 1. local training (honest nodes only; compromised stay frozen)
+        t_train0 = _time.perf_counter()
         with self.timer.phase("train"):
             if not self._is_compromised(self.rank):
                 self.node.local_train(
                     epochs=cfg.training.local_epochs, lr=cfg.training.lr,
                     round_num=round_num,
                 )
+        train_seconds = _time.perf_counter() - t_train0
+        ready = self._ready_mask(train_seconds)
+        if ready is not None:
+            topo = _drop_stragglers(topo, ready)
+            if not ready[self.rank]:
+                # overran the budget: skip exchange, keep own state
+                return
         # 2. snapshot + self-attack
         with self.timer.phase("snapshot_attack"):
             own = self.node.get_state()
@@ -167,6 +207,7 @@ class FLRoundLoop:
             and self.mobility is None
             and self._fully_connected
             and self.attack is None
+            and (ready is None or all(ready))  # a straggler skipped: no collective
         )
         if use_allreduce:
             # K1 folded into the collective: new state = global mean

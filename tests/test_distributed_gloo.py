@@ -131,3 +131,22 @@ def test_sketch_wire_mode_matches_full_exchange():
     for key in ["mean_accuracy", "mean_loss", "honest_accuracy"]:
         for a, b in zip(h_wire[key], h_full[key]):
             assert a == pytest.approx(b, abs=2e-3), (key, h_wire[key], h_full[key])
+
+
+def test_round_budget_straggler_semantics(monkeypatch):
+    """With a round budget, an overrunning node skips the exchange and its
+    peers aggregate without it (reference: node_process.py:210-217)."""
+    # budget so tiny every node is a "straggler": all skip exchange ->
+    # states evolve by local training only, never mixed
+    cfg = _base_config(2, algo="fedavg", topo="ring", rounds=2)
+    cfg["distributed"]["round_duration_s"] = 1e-9
+    h = _run_distributed(cfg, 2, 29630)
+    assert len(h["round"]) == 2
+    # and a generous budget behaves like no budget at all
+    cfg2 = _base_config(2, algo="fedavg", topo="ring", rounds=2)
+    cfg2["distributed"]["round_duration_s"] = 3600.0
+    h2 = _run_distributed(cfg2, 2, 29631)
+    base = _base_config(2, algo="fedavg", topo="ring", rounds=2)
+    h0 = _run_distributed(base, 2, 29632)
+    for a, b in zip(h2["mean_accuracy"], h0["mean_accuracy"]):
+        assert a == pytest.approx(b, abs=2e-3)
