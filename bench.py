@@ -155,6 +155,9 @@ def main():
     for r in range(args.warmup):
         one_round(r)
     sync()
+    # timing means should reflect steady state, not capture/comm-init rounds
+    loop.timer.totals.clear()
+    loop.timer.counts.clear()
     t0 = time.perf_counter()
     for r in range(args.warmup, args.warmup + args.steps):
         one_round(r)

@@ -149,8 +149,11 @@ class Node:
             if self._eval_graph is None:
                 from murmura_amd.core.gpu_round import EvalGraph
 
-                bs = getattr(self.test_loader, "batch_size", None) or 64
-                self._eval_graph = EvalGraph(self, shard, bs, self.evidential)
+                # evaluation is pure inference: accuracy/loss sums are
+                # batch-size independent (BN eval mode uses running stats),
+                # so batch as large as the shard allows to fill the 256 CUs
+                bs = min(shard.n, 1024)
+                self._eval_graph = EvalGraph(self, shard, max(1, bs), self.evidential)
             return self._eval_graph.run()
         if self.evidential:
             return evaluate_evidential(self.model, self.test_loader, self.device, self.dtype)
