@@ -5,6 +5,14 @@ from murmura_amd.aggregation.balance import BALANCEAggregator
 from murmura_amd.aggregation.sketchguard import SketchguardAggregator
 from murmura_amd.aggregation.ubar import UBARAggregator
 from murmura_amd.aggregation.evidential_trust import EvidentialTrustAggregator
+from murmura_amd.aggregation.compat import (
+    average_states,
+    calculate_model_dimension,
+    compute_model_distance,
+    flatten_model_state,
+    get_model_state,
+    set_model_state,
+)
 
 ALGORITHMS = {
     "fedavg": FedAvgAggregator,
@@ -27,4 +35,10 @@ __all__ = [
     "UBARAggregator",
     "EvidentialTrustAggregator",
     "ALGORITHMS",
+    "average_states",
+    "compute_model_distance",
+    "flatten_model_state",
+    "calculate_model_dimension",
+    "get_model_state",
+    "set_model_state",
 ]

@@ -352,6 +352,13 @@ def run_node_process(
                     f"honest={history['honest_accuracy'][-1]:.4f}",
                     flush=True,
                 )
+    # aggregate per-node aggregator statistics on rank 0 (the distributed
+    # analogue of Network.get_node_statistics)
+    stats = loop.node.aggregator.get_statistics()
+    stats["phase_timings_ms"] = loop.timer.summary()
+    all_stats = exchange.gather_metrics(stats, dst=0)
+    if rank == 0 and history is not None:
+        history["node_statistics"] = {i: s for i, s in enumerate(all_stats)}
     dist.barrier()
     if destroy_group:
         dist.destroy_process_group()

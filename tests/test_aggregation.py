@@ -237,3 +237,29 @@ def test_evidential_trust_ema():
     t2 = agg.get_statistics()["per_neighbor_trust"][1]
     # EMA: same raw trust each round keeps value roughly stable
     assert abs(t2 - t1) < 0.5
+
+
+# --------------------------------------------------------------- dict compat
+def test_compat_average_states_copies_nonfloat():
+    import torch.nn as nn
+
+    from murmura_amd.aggregation.compat import (
+        average_states,
+        compute_model_distance,
+        flatten_model_state,
+    )
+
+    m = nn.Sequential(nn.Linear(4, 4), nn.BatchNorm1d(4))
+    m(torch.randn(8, 4))  # tick num_batches_tracked
+    s1 = {k: v.clone() for k, v in m.state_dict().items()}
+    s2 = {k: v.clone() * 3 if torch.is_floating_point(v) else v.clone()
+          for k, v in s1.items()}
+    avg = average_states([s1, s2])
+    assert torch.allclose(avg["0.weight"], 2 * s1["0.weight"])
+    # non-float buffer copied, not averaged
+    assert avg["1.num_batches_tracked"].dtype == torch.int64
+    assert avg["1.num_batches_tracked"].item() == s1["1.num_batches_tracked"].item()
+    # distance and flatten agree
+    d = compute_model_distance(s1, s2)
+    f1, f2 = flatten_model_state(s1), flatten_model_state(s2)
+    assert d == pytest.approx((f1 - f2).norm().item(), rel=1e-5)
