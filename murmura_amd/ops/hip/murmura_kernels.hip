@@ -166,6 +166,64 @@ __global__ void gram_kernel(const T* __restrict__ x, int64_t P, float* __restric
   for (int t = threadIdx.x; t < NPAIR; t += blockDim.x) atomicAdd(&gram[t], lacc[t]);
 }
 
+// Cross-Gram tile: out[i, j] += <xi_i, xj_j> for an MI x MJ row-tile pair.
+// Used to decompose m > 10 Gram matrices into 8x8 tiles: the monolithic
+// kernel's m*(m+1)/2 accumulators spill registers at m ~ 12 (measured
+// 647 GB/s vs ~5600 at m=8); 8x8 tiles keep 64 accumulators + 2x8 row
+// packs in registers at the cost of re-reading rows once per tile pair.
+template <typename T, int MI, int MJ>
+__global__ void gram_cross_kernel(const T* __restrict__ xi, const T* __restrict__ xj,
+                                  int64_t P, float* __restrict__ out, int ldo) {
+  constexpr int N = Pack16<T>::N;
+  __shared__ float lacc[MI * MJ];
+  for (int t = threadIdx.x; t < MI * MJ; t += blockDim.x) lacc[t] = 0.0f;
+  __syncthreads();
+
+  float acc[MI * MJ];
+#pragma unroll
+  for (int t = 0; t < MI * MJ; ++t) acc[t] = 0.0f;
+
+  const int64_t nvec = P / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    Pack16<T> ri[MI], rj[MJ];
+#pragma unroll
+    for (int i = 0; i < MI; ++i) ri[i] = reinterpret_cast<const Pack16<T>*>(xi + (int64_t)i * P)[v];
+#pragma unroll
+    for (int j = 0; j < MJ; ++j) rj[j] = reinterpret_cast<const Pack16<T>*>(xj + (int64_t)j * P)[v];
+#pragma unroll
+    for (int i = 0; i < MI; ++i)
+#pragma unroll
+      for (int j = 0; j < MJ; ++j) {
+        float s = acc[i * MJ + j];
+#pragma unroll
+        for (int k = 0; k < N; ++k) s = fmaf(to_f(ri[i].e[k]), to_f(rj[j].e[k]), s);
+        acc[i * MJ + j] = s;
+      }
+  }
+  for (int64_t p = nvec * N + blockIdx.x * blockDim.x + threadIdx.x; p < P; p += stride) {
+    float fi[MI], fj[MJ];
+#pragma unroll
+    for (int i = 0; i < MI; ++i) fi[i] = to_f(xi[(int64_t)i * P + p]);
+#pragma unroll
+    for (int j = 0; j < MJ; ++j) fj[j] = to_f(xj[(int64_t)j * P + p]);
+#pragma unroll
+    for (int i = 0; i < MI; ++i)
+#pragma unroll
+      for (int j = 0; j < MJ; ++j) acc[i * MJ + j] += fi[i] * fj[j];
+  }
+
+#pragma unroll
+  for (int t = 0; t < MI * MJ; ++t) {
+    float s = wave_sum(acc[t]);
+    if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(&lacc[t], s);
+  }
+  __syncthreads();
+  for (int t = threadIdx.x; t < MI * MJ; t += blockDim.x) {
+    atomicAdd(&out[(t / MJ) * ldo + (t % MJ)], lacc[t]);
+  }
+}
+
 // ================================================================== K12
 // out[i] += sum_p x[i,p]^2  (per-row squared norms)
 template <typename T>
@@ -560,17 +618,61 @@ void launch_gram(const Tensor& x, Tensor& gram_flat, int64_t P) {
       (const elem_t*)x.data_ptr(), P, gram_flat.data_ptr<float>());
 }
 
+// tile-pair dispatch for the decomposed path: sizes are 8 or the ragged
+// remainder r = m % 8; only (8,8), (8,r) and (r,r) combinations occur.
+template <typename elem_t>
+void launch_gram_cross(const elem_t* xi, const elem_t* xj, int mi, int mj,
+                       int64_t P, float* out, int ldo) {
+  int blocks = grid_for(P / 4, BLOCK, 2048);
+  hipStream_t st = cur_stream();
+#define CROSS_CASE(MI, MJ)                                                       \
+  if (mi == MI && mj == MJ) {                                                    \
+    gram_cross_kernel<elem_t, MI, MJ>                                            \
+        <<<blocks, BLOCK, 0, st>>>(xi, xj, P, out, ldo);                         \
+    return;                                                                      \
+  }
+  CROSS_CASE(8, 8)
+  CROSS_CASE(8, 1) CROSS_CASE(8, 2) CROSS_CASE(8, 3) CROSS_CASE(8, 4)
+  CROSS_CASE(8, 5) CROSS_CASE(8, 6) CROSS_CASE(8, 7)
+  CROSS_CASE(1, 1) CROSS_CASE(2, 2) CROSS_CASE(3, 3) CROSS_CASE(4, 4)
+  CROSS_CASE(5, 5) CROSS_CASE(6, 6) CROSS_CASE(7, 7)
+#undef CROSS_CASE
+  TORCH_CHECK(false, "gram_cross: unsupported tile sizes ", mi, "x", mj);
+}
+
 Tensor pairwise_sq_dists(Tensor stacked) {
   check_flat(stacked, "stacked");
   TORCH_CHECK(stacked.dim() == 2);
   int m = (int)stacked.size(0);
   int64_t P = stacked.size(1);
-  if (m > 16) {
-    // plain library GEMM path for large m
+  if (m > 64) {
+    // plain library GEMM path for very large m
     Tensor x = stacked.to(at::kFloat);
     Tensor g = at::matmul(x, x.t());
     Tensor sq = g.diagonal();
     return (sq.unsqueeze(0) + sq.unsqueeze(1) - 2.0 * g).clamp_min_(0.0);
+  }
+  if (m > 10) {
+    // 8x8 row-tile decomposition (monolithic kernel spills registers here)
+    Tensor gram = at::zeros({m, m}, stacked.options().dtype(at::kFloat));
+    float* gp = gram.data_ptr<float>();
+    int ntiles = (m + 7) / 8;
+    DISPATCH_FT(stacked, {
+      const elem_t* base = (const elem_t*)stacked.data_ptr();
+      for (int a = 0; a < ntiles; ++a) {
+        int mi = std::min(8, m - a * 8);
+        for (int b = a; b < ntiles; ++b) {
+          int mj = std::min(8, m - b * 8);
+          launch_gram_cross<elem_t>(base + (int64_t)a * 8 * P,
+                                    base + (int64_t)b * 8 * P, mi, mj, P,
+                                    gp + (int64_t)a * 8 * m + b * 8, m);
+        }
+      }
+    });
+    // mirror the strict upper tiles into the lower triangle
+    Tensor full = gram.triu() + gram.triu(1).t();
+    Tensor sq = full.diagonal();
+    return (sq.unsqueeze(0) + sq.unsqueeze(1) - 2.0 * full).clamp_min_(0.0);
   }
   int npair = m * (m + 1) / 2;
   Tensor gram_flat = at::zeros({npair}, stacked.options().dtype(at::kFloat));

@@ -46,8 +46,9 @@ def test_pairwise_sq_dists(dtype, m):
     assert torch.all(d2.diagonal().abs() / expect.max().clamp_min(1e-6) < 1e-4)
 
 
-def test_pairwise_large_m_fallback():
-    x = _rand((20, 10_001))
+@pytest.mark.parametrize("m", [11, 13, 20, 33, 64, 65])
+def test_pairwise_tiled_and_fallback(m):
+    x = _rand((m, 10_001), seed=m)
     d2 = ops.pairwise_sq_dists(x)
     expect = ref.pairwise_sq_dists(x.float().cpu())
     assert torch.allclose(d2.cpu(), expect, rtol=1e-3, atol=1e-2)
