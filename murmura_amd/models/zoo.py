@@ -13,6 +13,8 @@ from typing import Callable, Dict
 import torch
 from torch import Tensor, nn
 
+from murmura_amd.ops.fused_bn import MurmuraBatchNorm2d
+
 
 class SimpleMLP(nn.Module):
     """Tiny MLP for synthetic-data tests (the reference's programmatic example
@@ -105,14 +107,14 @@ class BasicBlock(nn.Module):
     def __init__(self, in_ch: int, out_ch: int, stride: int = 1):
         super().__init__()
         self.conv1 = nn.Conv2d(in_ch, out_ch, 3, stride=stride, padding=1, bias=False)
-        self.bn1 = nn.BatchNorm2d(out_ch)
+        self.bn1 = MurmuraBatchNorm2d(out_ch)
         self.conv2 = nn.Conv2d(out_ch, out_ch, 3, stride=1, padding=1, bias=False)
-        self.bn2 = nn.BatchNorm2d(out_ch)
+        self.bn2 = MurmuraBatchNorm2d(out_ch)
         self.shortcut = nn.Sequential()
         if stride != 1 or in_ch != out_ch:
             self.shortcut = nn.Sequential(
                 nn.Conv2d(in_ch, out_ch, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(out_ch),
+                MurmuraBatchNorm2d(out_ch),
             )
 
     def forward(self, x: Tensor) -> Tensor:
@@ -128,7 +130,7 @@ class ResNet18(nn.Module):
     def __init__(self, num_classes: int = 10, in_channels: int = 3):
         super().__init__()
         self.conv1 = nn.Conv2d(in_channels, 64, 3, stride=1, padding=1, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = MurmuraBatchNorm2d(64)
         layers = []
         in_ch = 64
         for out_ch, stride in [(64, 1), (64, 1), (128, 2), (128, 1),

@@ -1,0 +1,70 @@
+"""Fused NHWC BatchNorm2d (K14).
+
+torch's channels-last BatchNorm kernels measured ~230 GB/s on MI355X
+(37 us per 8 MB channel reduction) and accounted for ~half of a ResNet-18
+bs-64 training step; the HIP kernels behind this module are plain two-pass
+streaming reductions with per-thread channel ownership (registers, not LDS
+atomics) targeting the HBM roofline.
+
+``MurmuraBatchNorm2d`` subclasses ``nn.BatchNorm2d`` — identical parameters,
+buffers, state-dict names and CPU/eager behavior; the fused path engages on
+ROCm for channels_last inputs with supported channel counts (C % 8 == 0 and
+C | 256 or 256 | C, C <= 1024).
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import Tensor, nn
+
+
+def _ext():
+    from murmura_amd.ops import _load_ext
+
+    return _load_ext()
+
+
+def _supported_c(c: int) -> bool:
+    return c % 8 == 0 and c <= 1024 and (256 % c == 0 or c % 256 == 0)
+
+
+class _FusedBNTrain(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, running_mean, running_var, momentum, eps):
+        y, mean, invstd = _ext().bn_fwd_train(
+            x, weight, bias, running_mean, running_var, momentum, eps
+        )
+        ctx.save_for_backward(x, weight, mean, invstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, invstd = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx, dweight, dbias = _ext().bn_bwd(x, dy, weight, mean, invstd)
+        return dx, dweight, dbias, None, None, None, None
+
+
+class MurmuraBatchNorm2d(nn.BatchNorm2d):
+    def forward(self, x: Tensor) -> Tensor:
+        use_fused = (
+            x.is_cuda
+            and x.dim() == 4
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and _supported_c(x.shape[1])
+            and self.track_running_stats
+            and self.affine
+            and _ext() is not None
+        )
+        if not use_fused:
+            return super().forward(x)
+        if self.training:
+            if self.num_batches_tracked is not None:
+                self.num_batches_tracked.add_(1)
+            return _FusedBNTrain.apply(
+                x, self.weight, self.bias, self.running_mean, self.running_var,
+                self.momentum if self.momentum is not None else 0.1, self.eps,
+            )
+        return _ext().bn_fwd_eval(
+            x, self.weight, self.bias, self.running_mean, self.running_var, self.eps
+        )

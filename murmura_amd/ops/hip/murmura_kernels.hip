@@ -565,6 +565,247 @@ __global__ void evidential_stats_kernel(const T* __restrict__ logits,
   }
 }
 
+// ================================================================== K14
+// Fused NHWC BatchNorm2d for the training hot path. torch's channels-last BN
+// kernels measured ~230 GB/s on MI355X (37 us for an 8 MB reduction);
+// these are plain two-pass streaming kernels targeting the HBM roofline.
+//
+// Memory view: x is [R, C] with C contiguous (R = N*H*W), C in {8..1024},
+// C % 8 == 0. fp32 accumulation/statistics regardless of element type.
+
+// pass 1: per-channel sum and sum-of-squares.
+// Each thread owns NC channels (c, c+stride, ...) so accumulation is pure
+// registers; one global atomic per channel per block.
+template <typename T, int NC>
+__global__ void bn_sums_kernel(const T* __restrict__ x, int64_t R, int C,
+                               float* __restrict__ sum, float* __restrict__ sumsq) {
+  const int cstride = blockDim.x;  // threads map to channels mod blockDim
+  const int c0 = threadIdx.x % C;  // valid when C <= blockDim
+  float s[NC], q[NC];
+#pragma unroll
+  for (int k = 0; k < NC; ++k) s[k] = q[k] = 0.0f;
+
+  if (C <= (int)blockDim.x) {
+    // groups of (blockDim/C) threads split the rows of one channel
+    const int rows_per_iter = blockDim.x / C;
+    const int rsub = threadIdx.x / C;
+    for (int64_t r = (int64_t)blockIdx.x * rows_per_iter + rsub; r < R;
+         r += (int64_t)gridDim.x * rows_per_iter) {
+      float v = to_f(x[r * C + c0]);
+      s[0] += v;
+      q[0] = fmaf(v, v, q[0]);
+    }
+    float partial[2] = {s[0], q[0]};
+    // combine the rows_per_iter threads sharing this channel via LDS
+    __shared__ float lsum[1024], lsq[1024];
+    lsum[threadIdx.x] = partial[0];
+    lsq[threadIdx.x] = partial[1];
+    __syncthreads();
+    if (threadIdx.x < (unsigned)C) {
+      float ts = 0.0f, tq = 0.0f;
+      for (int k = threadIdx.x; k < (int)blockDim.x; k += C) {
+        ts += lsum[k];
+        tq += lsq[k];
+      }
+      atomicAdd(&sum[threadIdx.x], ts);
+      atomicAdd(&sumsq[threadIdx.x], tq);
+    }
+  } else {
+    // C > blockDim: each thread owns NC = C/blockDim channels
+    for (int64_t r = blockIdx.x; r < R; r += gridDim.x) {
+#pragma unroll
+      for (int k = 0; k < NC; ++k) {
+        float v = to_f(x[r * C + c0 + k * cstride]);
+        s[k] += v;
+        q[k] = fmaf(v, v, q[k]);
+      }
+    }
+#pragma unroll
+    for (int k = 0; k < NC; ++k) {
+      atomicAdd(&sum[c0 + k * cstride], s[k]);
+      atomicAdd(&sumsq[c0 + k * cstride], q[k]);
+    }
+  }
+}
+
+// finalize: mean/invstd from sums; EMA-update running stats (torch semantics:
+// running_var uses the unbiased estimator)
+template <typename Tr>
+__global__ void bn_finalize_kernel(const float* __restrict__ sum,
+                                   const float* __restrict__ sumsq, int C, int64_t R,
+                                   float eps, float momentum,
+                                   float* __restrict__ mean, float* __restrict__ invstd,
+                                   Tr* __restrict__ running_mean,
+                                   Tr* __restrict__ running_var) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float m = sum[c] / (float)R;
+  float var = fmaxf(sumsq[c] / (float)R - m * m, 0.0f);
+  mean[c] = m;
+  invstd[c] = rsqrtf(var + eps);
+  if (running_mean != nullptr) {
+    from_f(running_mean[c],
+           (1.0f - momentum) * to_f(running_mean[c]) + momentum * m);
+    float unbiased = R > 1 ? var * (float)R / (float)(R - 1) : var;
+    from_f(running_var[c],
+           (1.0f - momentum) * to_f(running_var[c]) + momentum * unbiased);
+  }
+}
+
+// pass 2: y = (x - mean) * invstd * w + b, 16-byte packs over channels
+// (C % Pack16<T>::N == 0); per-channel coefficients staged in LDS.
+template <typename T>
+__global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y, int64_t R,
+                               int C, const float* __restrict__ mean,
+                               const float* __restrict__ invstd,
+                               const T* __restrict__ w, const T* __restrict__ b) {
+  extern __shared__ float coef[];  // scale[C], shift[C]
+  float* scale = coef;
+  float* shift = coef + C;
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float sc = invstd[c] * (w ? to_f(w[c]) : 1.0f);
+    scale[c] = sc;
+    shift[c] = (b ? to_f(b[c]) : 0.0f) - mean[c] * sc;
+  }
+  __syncthreads();
+  constexpr int N = Pack16<T>::N;
+  const int64_t nvec = R * C / N;
+  const int cvec = C / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    Pack16<T> pv = reinterpret_cast<const Pack16<T>*>(x)[v];
+    const int cbase = (int)(v % cvec) * N;
+#pragma unroll
+    for (int k = 0; k < N; ++k)
+      from_f(pv.e[k], fmaf(to_f(pv.e[k]), scale[cbase + k], shift[cbase + k]));
+    reinterpret_cast<Pack16<T>*>(y)[v] = pv;
+  }
+}
+
+// eval-mode forward: same as bn_norm but coefficients from running stats
+template <typename T>
+__global__ void bn_eval_kernel(const T* __restrict__ x, T* __restrict__ y, int64_t R,
+                               int C, const T* __restrict__ running_mean,
+                               const T* __restrict__ running_var, float eps,
+                               const T* __restrict__ w, const T* __restrict__ b) {
+  extern __shared__ float coef[];
+  float* scale = coef;
+  float* shift = coef + C;
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float inv = rsqrtf(to_f(running_var[c]) + eps);
+    float sc = inv * (w ? to_f(w[c]) : 1.0f);
+    scale[c] = sc;
+    shift[c] = (b ? to_f(b[c]) : 0.0f) - to_f(running_mean[c]) * sc;
+  }
+  __syncthreads();
+  constexpr int N = Pack16<T>::N;
+  const int64_t nvec = R * C / N;
+  const int cvec = C / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    Pack16<T> pv = reinterpret_cast<const Pack16<T>*>(x)[v];
+    const int cbase = (int)(v % cvec) * N;
+#pragma unroll
+    for (int k = 0; k < N; ++k)
+      from_f(pv.e[k], fmaf(to_f(pv.e[k]), scale[cbase + k], shift[cbase + k]));
+    reinterpret_cast<Pack16<T>*>(y)[v] = pv;
+  }
+}
+
+// backward pass 1: per-channel sum(dy) and sum(dy * xhat)
+template <typename T, int NC>
+__global__ void bn_bwd_sums_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                                   int64_t R, int C, const float* __restrict__ mean,
+                                   const float* __restrict__ invstd,
+                                   float* __restrict__ sum_dy,
+                                   float* __restrict__ sum_dyx) {
+  const int cstride = blockDim.x;
+  const int c0 = threadIdx.x % C;
+  float s[NC], q[NC];
+#pragma unroll
+  for (int k = 0; k < NC; ++k) s[k] = q[k] = 0.0f;
+
+  if (C <= (int)blockDim.x) {
+    const int rows_per_iter = blockDim.x / C;
+    const int rsub = threadIdx.x / C;
+    const float m = mean[c0], is = invstd[c0];
+    for (int64_t r = (int64_t)blockIdx.x * rows_per_iter + rsub; r < R;
+         r += (int64_t)gridDim.x * rows_per_iter) {
+      float g = to_f(dy[r * C + c0]);
+      float xh = (to_f(x[r * C + c0]) - m) * is;
+      s[0] += g;
+      q[0] = fmaf(g, xh, q[0]);
+    }
+    __shared__ float lsum[1024], lsq[1024];
+    lsum[threadIdx.x] = s[0];
+    lsq[threadIdx.x] = q[0];
+    __syncthreads();
+    if (threadIdx.x < (unsigned)C) {
+      float ts = 0.0f, tq = 0.0f;
+      for (int k = threadIdx.x; k < (int)blockDim.x; k += C) {
+        ts += lsum[k];
+        tq += lsq[k];
+      }
+      atomicAdd(&sum_dy[threadIdx.x], ts);
+      atomicAdd(&sum_dyx[threadIdx.x], tq);
+    }
+  } else {
+    for (int64_t r = blockIdx.x; r < R; r += gridDim.x) {
+#pragma unroll
+      for (int k = 0; k < NC; ++k) {
+        const int c = c0 + k * cstride;
+        float g = to_f(dy[r * C + c]);
+        float xh = (to_f(x[r * C + c]) - mean[c]) * invstd[c];
+        s[k] += g;
+        q[k] = fmaf(g, xh, q[k]);
+      }
+    }
+#pragma unroll
+    for (int k = 0; k < NC; ++k) {
+      atomicAdd(&sum_dy[c0 + k * cstride], s[k]);
+      atomicAdd(&sum_dyx[c0 + k * cstride], q[k]);
+    }
+  }
+}
+
+// backward pass 2: dx = w*invstd * (dy - sum_dy/R - xhat * sum_dyx/R)
+template <typename T>
+__global__ void bn_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                                 T* __restrict__ dx, int64_t R, int C,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ invstd,
+                                 const T* __restrict__ w,
+                                 const float* __restrict__ sum_dy,
+                                 const float* __restrict__ sum_dyx) {
+  extern __shared__ float coef[];  // g1[C], g2[C], g3[C]
+  float* g_scale = coef;           // w*invstd
+  float* g_mean = coef + C;        // sum_dy / R
+  float* g_proj = coef + 2 * C;    // sum_dyx / R * invstd (per xhat term)
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    g_scale[c] = (w ? to_f(w[c]) : 1.0f) * invstd[c];
+    g_mean[c] = sum_dy[c] / (float)R;
+    g_proj[c] = sum_dyx[c] / (float)R;
+  }
+  __syncthreads();
+  constexpr int N = Pack16<T>::N;
+  const int64_t nvec = R * C / N;
+  const int cvec = C / N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+    Pack16<T> xv = reinterpret_cast<const Pack16<T>*>(x)[v];
+    Pack16<T> gv = reinterpret_cast<const Pack16<T>*>(dy)[v];
+    const int cbase = (int)(v % cvec) * N;
+#pragma unroll
+    for (int k = 0; k < N; ++k) {
+      const int c = cbase + k;
+      float xh = (to_f(xv.e[k]) - mean[c]) * invstd[c];
+      float t = to_f(gv.e[k]) - g_mean[c] - xh * g_proj[c];
+      from_f(gv.e[k], t * g_scale[c]);
+    }
+    reinterpret_cast<Pack16<T>*>(dx)[v] = gv;
+  }
+}
+
 // =================================================================
 // bindings
 // =================================================================
@@ -846,9 +1087,137 @@ Tensor evidential_stats(Tensor logits, Tensor targets) {
   return out;
 }
 
+// ---------------------------------------------------------------- K14 bindings
+// x: channels_last 4-D tensor (underlying memory [N*H*W, C] with C contiguous)
+static int64_t bn_check(const Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "bn: x must be 4-D cuda");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "bn: x must be channels_last");
+  int C = (int)x.size(1);
+  TORCH_CHECK(C % 8 == 0 && (256 % C == 0 || C % 256 == 0) && C <= 1024,
+              "bn: unsupported channel count ", C);
+  return x.numel() / C;
+}
+
+template <typename elem_t>
+void bn_sums_dispatch(const Tensor& x, int64_t R, int C, Tensor& sum, Tensor& sumsq) {
+  int blocks = grid_for(R, 4, 2048);
+  hipStream_t st = cur_stream();
+  if (C <= 256) {
+    bn_sums_kernel<elem_t, 1><<<blocks, BLOCK, 0, st>>>(
+        (const elem_t*)x.data_ptr(), R, C, sum.data_ptr<float>(), sumsq.data_ptr<float>());
+  } else if (C == 512) {
+    bn_sums_kernel<elem_t, 2><<<blocks, BLOCK, 0, st>>>(
+        (const elem_t*)x.data_ptr(), R, C, sum.data_ptr<float>(), sumsq.data_ptr<float>());
+  } else {
+    bn_sums_kernel<elem_t, 4><<<blocks, BLOCK, 0, st>>>(
+        (const elem_t*)x.data_ptr(), R, C, sum.data_ptr<float>(), sumsq.data_ptr<float>());
+  }
+}
+
+std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optional<Tensor> b,
+                                 c10::optional<Tensor> running_mean,
+                                 c10::optional<Tensor> running_var, double momentum,
+                                 double eps) {
+  int64_t R = bn_check(x);
+  int C = (int)x.size(1);
+  auto fopt = x.options().dtype(at::kFloat);
+  Tensor sum = at::zeros({C}, fopt), sumsq = at::zeros({C}, fopt);
+  Tensor mean = at::empty({C}, fopt), invstd = at::empty({C}, fopt);
+  DISPATCH_FT(x, { bn_sums_dispatch<elem_t>(x, R, C, sum, sumsq); });
+  if (running_mean.has_value()) {
+    TORCH_CHECK(running_mean->scalar_type() == x.scalar_type(),
+                "bn: running stats must match input dtype");
+  }
+  DISPATCH_FT(x, {
+    elem_t* rm = running_mean.has_value() ? (elem_t*)running_mean->data_ptr() : nullptr;
+    elem_t* rv = running_var.has_value() ? (elem_t*)running_var->data_ptr() : nullptr;
+    bn_finalize_kernel<elem_t><<<(C + 255) / 256, 256, 0, cur_stream()>>>(
+        sum.data_ptr<float>(), sumsq.data_ptr<float>(), C, R, (float)eps,
+        (float)momentum, mean.data_ptr<float>(), invstd.data_ptr<float>(), rm, rv);
+  });
+  Tensor y = at::empty_like(x);
+  int blocks = grid_for(R * C / 4, BLOCK);
+  size_t lds = 2 * C * sizeof(float);
+  DISPATCH_FT(x, {
+    bn_norm_kernel<elem_t><<<blocks, BLOCK, lds, cur_stream()>>>(
+        (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
+        mean.data_ptr<float>(), invstd.data_ptr<float>(),
+        w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+        b.has_value() ? (const elem_t*)b->data_ptr() : nullptr);
+  });
+  return {y, mean, invstd};
+}
+
+Tensor bn_fwd_eval(Tensor x, c10::optional<Tensor> w, c10::optional<Tensor> b,
+                   Tensor running_mean, Tensor running_var, double eps) {
+  int64_t R = bn_check(x);
+  int C = (int)x.size(1);
+  Tensor y = at::empty_like(x);
+  int blocks = grid_for(R * C / 4, BLOCK);
+  size_t lds = 2 * C * sizeof(float);
+  TORCH_CHECK(running_mean.scalar_type() == x.scalar_type(),
+              "bn: running stats must match input dtype");
+  DISPATCH_FT(x, {
+    bn_eval_kernel<elem_t><<<blocks, BLOCK, lds, cur_stream()>>>(
+        (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
+        (const elem_t*)running_mean.data_ptr(),
+        (const elem_t*)running_var.data_ptr(), (float)eps,
+        w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+        b.has_value() ? (const elem_t*)b->data_ptr() : nullptr);
+  });
+  return y;
+}
+
+std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor mean,
+                           Tensor invstd) {
+  int64_t R = bn_check(x);
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast));
+  int C = (int)x.size(1);
+  auto fopt = x.options().dtype(at::kFloat);
+  Tensor sum_dy = at::zeros({C}, fopt), sum_dyx = at::zeros({C}, fopt);
+  hipStream_t st = cur_stream();
+  DISPATCH_FT(x, {
+    int blocks = grid_for(R, 4, 2048);
+    if (C <= 256) {
+      bn_bwd_sums_kernel<elem_t, 1><<<blocks, BLOCK, 0, st>>>(
+          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), R, C,
+          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
+    } else if (C == 512) {
+      bn_bwd_sums_kernel<elem_t, 2><<<blocks, BLOCK, 0, st>>>(
+          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), R, C,
+          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
+    } else {
+      bn_bwd_sums_kernel<elem_t, 4><<<blocks, BLOCK, 0, st>>>(
+          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), R, C,
+          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
+    }
+  });
+  Tensor dx = at::empty_like(dy);
+  int blocks = grid_for(R * C / 4, BLOCK);
+  size_t lds = 3 * C * sizeof(float);
+  DISPATCH_FT(x, {
+    bn_bwd_dx_kernel<elem_t><<<blocks, BLOCK, lds, st>>>(
+        (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(),
+        (elem_t*)dx.data_ptr(), R, C, mean.data_ptr<float>(),
+        invstd.data_ptr<float>(),
+        w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
+  });
+  // dweight = sum_dyx (in input dtype), dbias = sum_dy
+  return {dx, sum_dyx.to(x.scalar_type()), sum_dy.to(x.scalar_type())};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bn_fwd_train", &bn_fwd_train,
+        "K14: fused NHWC BatchNorm training forward -> (y, mean, invstd)");
+  m.def("bn_fwd_eval", &bn_fwd_eval, "K14: NHWC BatchNorm eval forward");
+  m.def("bn_bwd", &bn_bwd, "K14: NHWC BatchNorm backward -> (dx, dweight, dbias)");
   m.def("weighted_sum", &weighted_sum, "K1: out = sum_i w_i * x_i",
         py::arg("stacked"), py::arg("w"), py::arg("out") = py::none());
   m.def("pairwise_sq_dists", &pairwise_sq_dists, "K2: [m,m] squared L2 matrix");

@@ -3,6 +3,7 @@
 
 import pytest
 import torch
+from torch import nn
 
 pytestmark = pytest.mark.gpu
 
@@ -181,3 +182,72 @@ def test_native_required_on_gpu(monkeypatch):
     x = torch.randn(10, device="cuda")
     with pytest.raises(RuntimeError, match="HIP extension"):
         O.weighted_sum(x.view(1, -1).contiguous(), torch.ones(1, device="cuda"))
+
+
+# ---------------------------------------------------------------- K14 fused BN
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("c", [64, 128, 256, 512])
+def test_fused_bn_forward_backward_matches_torch(dtype, c):
+    from murmura_amd.ops.fused_bn import MurmuraBatchNorm2d
+
+    torch.manual_seed(c)
+    n, h, w = 16, 8, 8
+    x = torch.randn(n, c, h, w, device="cuda", dtype=dtype)
+    xf = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    xr = x.clone().float().requires_grad_(True)
+
+    fused = MurmuraBatchNorm2d(c).cuda().to(dtype).to(memory_format=torch.channels_last)
+    refbn = nn.BatchNorm2d(c).cuda().float()
+    with torch.no_grad():
+        refbn.weight.copy_(fused.weight.float())
+        refbn.bias.copy_(fused.bias.float())
+
+    fused.train()
+    refbn.train()
+    y = fused(xf)
+    yr = refbn(xr)
+    tol = 2e-5 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(y.float(), yr, atol=tol, rtol=tol)
+    # running stats updated identically (unbiased var, momentum 0.1)
+    assert torch.allclose(fused.running_mean.float(), refbn.running_mean, atol=tol, rtol=1e-2)
+    assert torch.allclose(fused.running_var.float(), refbn.running_var, atol=tol, rtol=1e-2)
+    assert fused.num_batches_tracked.item() == 1
+
+    g = torch.randn_like(yr)
+    y.backward(g.to(dtype).contiguous(memory_format=torch.channels_last))
+    yr.backward(g)
+    gtol = 1e-4 if dtype == torch.float32 else 6e-2
+    scale = xr.grad.abs().max().clamp_min(1e-6)
+    assert ((xf.grad.float() - xr.grad).abs().max() / scale).item() < gtol
+    wscale = refbn.weight.grad.abs().max().clamp_min(1e-6)
+    assert ((fused.weight.grad.float() - refbn.weight.grad).abs().max() / wscale).item() < gtol
+    assert ((fused.bias.grad.float() - refbn.bias.grad).abs().max() / wscale).item() < gtol
+
+
+def test_fused_bn_eval_matches_torch():
+    from murmura_amd.ops.fused_bn import MurmuraBatchNorm2d
+
+    torch.manual_seed(1)
+    c = 64
+    fused = MurmuraBatchNorm2d(c).cuda().to(memory_format=torch.channels_last)
+    with torch.no_grad():
+        fused.running_mean.uniform_(-1, 1)
+        fused.running_var.uniform_(0.5, 2.0)
+    refbn = nn.BatchNorm2d(c).cuda()
+    refbn.load_state_dict(fused.state_dict())
+    fused.eval()
+    refbn.eval()
+    x = torch.randn(8, c, 4, 4, device="cuda")
+    y = fused(x.contiguous(memory_format=torch.channels_last))
+    yr = refbn(x)
+    assert torch.allclose(y, yr, atol=1e-5, rtol=1e-5)
+
+
+def test_fused_bn_cpu_fallback_identical_to_torch():
+    from murmura_amd.ops.fused_bn import MurmuraBatchNorm2d
+
+    m = MurmuraBatchNorm2d(16)
+    r = nn.BatchNorm2d(16)
+    r.load_state_dict(m.state_dict())
+    x = torch.randn(4, 16, 5, 5)
+    assert torch.allclose(m(x), r(x))
