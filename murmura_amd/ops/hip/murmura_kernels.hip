@@ -574,57 +574,51 @@ __global__ void evidential_stats_kernel(const T* __restrict__ logits,
 // C % 8 == 0. fp32 accumulation/statistics regardless of element type.
 
 // pass 1: per-channel sum and sum-of-squares.
-// Each thread owns NC channels (c, c+stride, ...) so accumulation is pure
-// registers; one global atomic per channel per block.
-template <typename T, int NC>
+// Each thread owns one 16-byte CHANNEL PACK (8 bf16 / 4 fp32 channels) and
+// accumulates in registers while streaming rows; per-block combine goes
+// through one LDS histogram pass, then one global atomic per channel per
+// block. (The first version assigned one scalar channel per thread — 2-byte
+// loads, 190 GB/s; packs restore the streaming roofline.)
+template <typename T>
 __global__ void bn_sums_kernel(const T* __restrict__ x, int64_t R, int C,
                                float* __restrict__ sum, float* __restrict__ sumsq) {
-  const int cstride = blockDim.x;  // threads map to channels mod blockDim
-  const int c0 = threadIdx.x % C;  // valid when C <= blockDim
-  float s[NC], q[NC];
-#pragma unroll
-  for (int k = 0; k < NC; ++k) s[k] = q[k] = 0.0f;
+  constexpr int N = Pack16<T>::N;
+  const int packs_per_row = C / N;          // caller guarantees divisibility
+  const int pk = threadIdx.x % packs_per_row;
+  const int rsub = threadIdx.x / packs_per_row;
+  const int rows_per_iter = blockDim.x / packs_per_row;
+  const int cbase = pk * N;
 
-  if (C <= (int)blockDim.x) {
-    // groups of (blockDim/C) threads split the rows of one channel
-    const int rows_per_iter = blockDim.x / C;
-    const int rsub = threadIdx.x / C;
+  float sacc[N], qacc[N];
+#pragma unroll
+  for (int k = 0; k < N; ++k) sacc[k] = qacc[k] = 0.0f;
+
+  if (rsub < rows_per_iter) {
     for (int64_t r = (int64_t)blockIdx.x * rows_per_iter + rsub; r < R;
          r += (int64_t)gridDim.x * rows_per_iter) {
-      float v = to_f(x[r * C + c0]);
-      s[0] += v;
-      q[0] = fmaf(v, v, q[0]);
-    }
-    float partial[2] = {s[0], q[0]};
-    // combine the rows_per_iter threads sharing this channel via LDS
-    __shared__ float lsum[1024], lsq[1024];
-    lsum[threadIdx.x] = partial[0];
-    lsq[threadIdx.x] = partial[1];
-    __syncthreads();
-    if (threadIdx.x < (unsigned)C) {
-      float ts = 0.0f, tq = 0.0f;
-      for (int k = threadIdx.x; k < (int)blockDim.x; k += C) {
-        ts += lsum[k];
-        tq += lsq[k];
-      }
-      atomicAdd(&sum[threadIdx.x], ts);
-      atomicAdd(&sumsq[threadIdx.x], tq);
-    }
-  } else {
-    // C > blockDim: each thread owns NC = C/blockDim channels
-    for (int64_t r = blockIdx.x; r < R; r += gridDim.x) {
+      Pack16<T> v = *reinterpret_cast<const Pack16<T>*>(x + r * C + cbase);
 #pragma unroll
-      for (int k = 0; k < NC; ++k) {
-        float v = to_f(x[r * C + c0 + k * cstride]);
-        s[k] += v;
-        q[k] = fmaf(v, v, q[k]);
+      for (int k = 0; k < N; ++k) {
+        float f = to_f(v.e[k]);
+        sacc[k] += f;
+        qacc[k] = fmaf(f, f, qacc[k]);
       }
     }
+  }
+  extern __shared__ float lds[];  // csum[C], csq[C]
+  float* csum = lds;
+  float* csq = lds + C;
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.0f;
+  __syncthreads();
 #pragma unroll
-    for (int k = 0; k < NC; ++k) {
-      atomicAdd(&sum[c0 + k * cstride], s[k]);
-      atomicAdd(&sumsq[c0 + k * cstride], q[k]);
-    }
+  for (int k = 0; k < N; ++k) {
+    atomicAdd(&csum[cbase + k], sacc[k]);
+    atomicAdd(&csq[cbase + k], qacc[k]);
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    atomicAdd(&sum[c], csum[c]);
+    atomicAdd(&sumsq[c], csq[c]);
   }
 }
 
@@ -712,59 +706,57 @@ __global__ void bn_eval_kernel(const T* __restrict__ x, T* __restrict__ y, int64
   }
 }
 
-// backward pass 1: per-channel sum(dy) and sum(dy * xhat)
-template <typename T, int NC>
+// backward pass 1: per-channel sum(dy) and sum(dy * xhat), same packed
+// streaming structure as bn_sums_kernel.
+template <typename T>
 __global__ void bn_bwd_sums_kernel(const T* __restrict__ x, const T* __restrict__ dy,
                                    int64_t R, int C, const float* __restrict__ mean,
                                    const float* __restrict__ invstd,
                                    float* __restrict__ sum_dy,
                                    float* __restrict__ sum_dyx) {
-  const int cstride = blockDim.x;
-  const int c0 = threadIdx.x % C;
-  float s[NC], q[NC];
-#pragma unroll
-  for (int k = 0; k < NC; ++k) s[k] = q[k] = 0.0f;
+  constexpr int N = Pack16<T>::N;
+  const int packs_per_row = C / N;
+  const int pk = threadIdx.x % packs_per_row;
+  const int rsub = threadIdx.x / packs_per_row;
+  const int rows_per_iter = blockDim.x / packs_per_row;
+  const int cbase = pk * N;
 
-  if (C <= (int)blockDim.x) {
-    const int rows_per_iter = blockDim.x / C;
-    const int rsub = threadIdx.x / C;
-    const float m = mean[c0], is = invstd[c0];
+  float m[N], is[N], sacc[N], qacc[N];
+#pragma unroll
+  for (int k = 0; k < N; ++k) {
+    m[k] = mean[cbase + k];
+    is[k] = invstd[cbase + k];
+    sacc[k] = qacc[k] = 0.0f;
+  }
+
+  if (rsub < rows_per_iter) {
     for (int64_t r = (int64_t)blockIdx.x * rows_per_iter + rsub; r < R;
          r += (int64_t)gridDim.x * rows_per_iter) {
-      float g = to_f(dy[r * C + c0]);
-      float xh = (to_f(x[r * C + c0]) - m) * is;
-      s[0] += g;
-      q[0] = fmaf(g, xh, q[0]);
-    }
-    __shared__ float lsum[1024], lsq[1024];
-    lsum[threadIdx.x] = s[0];
-    lsq[threadIdx.x] = q[0];
-    __syncthreads();
-    if (threadIdx.x < (unsigned)C) {
-      float ts = 0.0f, tq = 0.0f;
-      for (int k = threadIdx.x; k < (int)blockDim.x; k += C) {
-        ts += lsum[k];
-        tq += lsq[k];
-      }
-      atomicAdd(&sum_dy[threadIdx.x], ts);
-      atomicAdd(&sum_dyx[threadIdx.x], tq);
-    }
-  } else {
-    for (int64_t r = blockIdx.x; r < R; r += gridDim.x) {
+      Pack16<T> xv = *reinterpret_cast<const Pack16<T>*>(x + r * C + cbase);
+      Pack16<T> gv = *reinterpret_cast<const Pack16<T>*>(dy + r * C + cbase);
 #pragma unroll
-      for (int k = 0; k < NC; ++k) {
-        const int c = c0 + k * cstride;
-        float g = to_f(dy[r * C + c]);
-        float xh = (to_f(x[r * C + c]) - mean[c]) * invstd[c];
-        s[k] += g;
-        q[k] = fmaf(g, xh, q[k]);
+      for (int k = 0; k < N; ++k) {
+        float g = to_f(gv.e[k]);
+        float xh = (to_f(xv.e[k]) - m[k]) * is[k];
+        sacc[k] += g;
+        qacc[k] = fmaf(g, xh, qacc[k]);
       }
     }
+  }
+  extern __shared__ float lds[];  // csum[C], csq[C]
+  float* csum = lds;
+  float* csq = lds + C;
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.0f;
+  __syncthreads();
 #pragma unroll
-    for (int k = 0; k < NC; ++k) {
-      atomicAdd(&sum_dy[c0 + k * cstride], s[k]);
-      atomicAdd(&sum_dyx[c0 + k * cstride], q[k]);
-    }
+  for (int k = 0; k < N; ++k) {
+    atomicAdd(&csum[cbase + k], sacc[k]);
+    atomicAdd(&csq[cbase + k], qacc[k]);
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    atomicAdd(&sum_dy[c], csum[c]);
+    atomicAdd(&sum_dyx[c], csq[c]);
   }
 }
 
@@ -1094,25 +1086,19 @@ static int64_t bn_check(const Tensor& x) {
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
               "bn: x must be channels_last");
   int C = (int)x.size(1);
-  TORCH_CHECK(C % 8 == 0 && (256 % C == 0 || C % 256 == 0) && C <= 1024,
-              "bn: unsupported channel count ", C);
+  TORCH_CHECK(C % 8 == 0 && C <= 2048, "bn: unsupported channel count ", C);
   return x.numel() / C;
 }
 
 template <typename elem_t>
 void bn_sums_dispatch(const Tensor& x, int64_t R, int C, Tensor& sum, Tensor& sumsq) {
-  int blocks = grid_for(R, 4, 2048);
-  hipStream_t st = cur_stream();
-  if (C <= 256) {
-    bn_sums_kernel<elem_t, 1><<<blocks, BLOCK, 0, st>>>(
-        (const elem_t*)x.data_ptr(), R, C, sum.data_ptr<float>(), sumsq.data_ptr<float>());
-  } else if (C == 512) {
-    bn_sums_kernel<elem_t, 2><<<blocks, BLOCK, 0, st>>>(
-        (const elem_t*)x.data_ptr(), R, C, sum.data_ptr<float>(), sumsq.data_ptr<float>());
-  } else {
-    bn_sums_kernel<elem_t, 4><<<blocks, BLOCK, 0, st>>>(
-        (const elem_t*)x.data_ptr(), R, C, sum.data_ptr<float>(), sumsq.data_ptr<float>());
-  }
+  constexpr int N = Pack16<elem_t>::N;
+  const int packs_per_row = C / N;
+  const int rows_per_iter = std::max(1, BLOCK / packs_per_row);
+  int blocks = grid_for(R, rows_per_iter, 2048);
+  size_t lds = 2 * C * sizeof(float);
+  bn_sums_kernel<elem_t><<<blocks, BLOCK, lds, cur_stream()>>>(
+      (const elem_t*)x.data_ptr(), R, C, sum.data_ptr<float>(), sumsq.data_ptr<float>());
 }
 
 std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optional<Tensor> b,
@@ -1178,23 +1164,15 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
   Tensor sum_dy = at::zeros({C}, fopt), sum_dyx = at::zeros({C}, fopt);
   hipStream_t st = cur_stream();
   DISPATCH_FT(x, {
-    int blocks = grid_for(R, 4, 2048);
-    if (C <= 256) {
-      bn_bwd_sums_kernel<elem_t, 1><<<blocks, BLOCK, 0, st>>>(
-          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), R, C,
-          mean.data_ptr<float>(), invstd.data_ptr<float>(),
-          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
-    } else if (C == 512) {
-      bn_bwd_sums_kernel<elem_t, 2><<<blocks, BLOCK, 0, st>>>(
-          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), R, C,
-          mean.data_ptr<float>(), invstd.data_ptr<float>(),
-          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
-    } else {
-      bn_bwd_sums_kernel<elem_t, 4><<<blocks, BLOCK, 0, st>>>(
-          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), R, C,
-          mean.data_ptr<float>(), invstd.data_ptr<float>(),
-          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
-    }
+    constexpr int N = Pack16<elem_t>::N;
+    const int packs_per_row = C / N;
+    const int rows_per_iter = std::max(1, BLOCK / packs_per_row);
+    int blocks = grid_for(R, rows_per_iter, 2048);
+    size_t lds = 2 * C * sizeof(float);
+    bn_bwd_sums_kernel<elem_t><<<blocks, BLOCK, lds, st>>>(
+        (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), R, C,
+        mean.data_ptr<float>(), invstd.data_ptr<float>(),
+        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
   });
   Tensor dx = at::empty_like(dy);
   int blocks = grid_for(R * C / 4, BLOCK);
