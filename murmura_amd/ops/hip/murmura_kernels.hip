@@ -622,44 +622,41 @@ __global__ void bn_sums_kernel(const T* __restrict__ x, int64_t R, int C,
   }
 }
 
-// finalize: mean/invstd from sums; EMA-update running stats (torch semantics:
-// running_var uses the unbiased estimator)
-template <typename Tr>
-__global__ void bn_finalize_kernel(const float* __restrict__ sum,
-                                   const float* __restrict__ sumsq, int C, int64_t R,
-                                   float eps, float momentum,
-                                   float* __restrict__ mean, float* __restrict__ invstd,
-                                   Tr* __restrict__ running_mean,
-                                   Tr* __restrict__ running_var) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float m = sum[c] / (float)R;
-  float var = fmaxf(sumsq[c] / (float)R - m * m, 0.0f);
-  mean[c] = m;
-  invstd[c] = rsqrtf(var + eps);
-  if (running_mean != nullptr) {
-    from_f(running_mean[c],
-           (1.0f - momentum) * to_f(running_mean[c]) + momentum * m);
-    float unbiased = R > 1 ? var * (float)R / (float)(R - 1) : var;
-    from_f(running_var[c],
-           (1.0f - momentum) * to_f(running_var[c]) + momentum * unbiased);
-  }
-}
-
-// pass 2: y = (x - mean) * invstd * w + b, 16-byte packs over channels
-// (C % Pack16<T>::N == 0); per-channel coefficients staged in LDS.
+// pass 2 (training): per-block recompute of mean/invstd from the sums
+// (C ops, cheap) -> normalize; block 0 additionally persists save_mean /
+// save_invstd for backward and EMA-updates the running stats. Folding the
+// old finalize kernel in here saves one launch per BN call (launch count is
+// what dominates small-batch BN inside graphs).
 template <typename T>
-__global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y, int64_t R,
-                               int C, const float* __restrict__ mean,
-                               const float* __restrict__ invstd,
-                               const T* __restrict__ w, const T* __restrict__ b) {
+__global__ void bn_norm_train_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                     int64_t R, int C,
+                                     const float* __restrict__ sum,
+                                     const float* __restrict__ sumsq, float eps,
+                                     float momentum, const T* __restrict__ w,
+                                     const T* __restrict__ b,
+                                     float* __restrict__ save_mean,
+                                     float* __restrict__ save_invstd,
+                                     T* __restrict__ running_mean,
+                                     T* __restrict__ running_var) {
   extern __shared__ float coef[];  // scale[C], shift[C]
   float* scale = coef;
   float* shift = coef + C;
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float sc = invstd[c] * (w ? to_f(w[c]) : 1.0f);
+    float m = sum[c] / (float)R;
+    float var = fmaxf(sumsq[c] / (float)R - m * m, 0.0f);
+    float inv = rsqrtf(var + eps);
+    float sc = inv * (w ? to_f(w[c]) : 1.0f);
     scale[c] = sc;
-    shift[c] = (b ? to_f(b[c]) : 0.0f) - mean[c] * sc;
+    shift[c] = (b ? to_f(b[c]) : 0.0f) - m * sc;
+    if (blockIdx.x == 0) {
+      save_mean[c] = m;
+      save_invstd[c] = inv;
+      if (running_mean != nullptr) {
+        from_f(running_mean[c], (1.0f - momentum) * to_f(running_mean[c]) + momentum * m);
+        float unbiased = R > 1 ? var * (float)R / (float)(R - 1) : var;
+        from_f(running_var[c], (1.0f - momentum) * to_f(running_var[c]) + momentum * unbiased);
+      }
+    }
   }
   __syncthreads();
   constexpr int N = Pack16<T>::N;
@@ -768,7 +765,8 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ 
                                  const float* __restrict__ invstd,
                                  const T* __restrict__ w,
                                  const float* __restrict__ sum_dy,
-                                 const float* __restrict__ sum_dyx) {
+                                 const float* __restrict__ sum_dyx,
+                                 T* __restrict__ dweight, T* __restrict__ dbias) {
   extern __shared__ float coef[];  // g1[C], g2[C], g3[C]
   float* g_scale = coef;           // w*invstd
   float* g_mean = coef + C;        // sum_dy / R
@@ -777,6 +775,10 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ 
     g_scale[c] = (w ? to_f(w[c]) : 1.0f) * invstd[c];
     g_mean[c] = sum_dy[c] / (float)R;
     g_proj[c] = sum_dyx[c] / (float)R;
+    if (blockIdx.x == 0) {
+      from_f(dweight[c], sum_dyx[c]);
+      from_f(dbias[c], sum_dy[c]);
+    }
   }
   __syncthreads();
   constexpr int N = Pack16<T>::N;
@@ -1108,31 +1110,29 @@ std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optiona
   int64_t R = bn_check(x);
   int C = (int)x.size(1);
   auto fopt = x.options().dtype(at::kFloat);
-  Tensor sum = at::zeros({C}, fopt), sumsq = at::zeros({C}, fopt);
-  Tensor mean = at::empty({C}, fopt), invstd = at::empty({C}, fopt);
+  Tensor ws = at::zeros({2, C}, fopt);     // one fill covers sum + sumsq
+  Tensor sum = ws[0], sumsq = ws[1];
+  Tensor saved = at::empty({2, C}, fopt);  // save_mean + save_invstd
   DISPATCH_FT(x, { bn_sums_dispatch<elem_t>(x, R, C, sum, sumsq); });
   if (running_mean.has_value()) {
     TORCH_CHECK(running_mean->scalar_type() == x.scalar_type(),
                 "bn: running stats must match input dtype");
   }
-  DISPATCH_FT(x, {
-    elem_t* rm = running_mean.has_value() ? (elem_t*)running_mean->data_ptr() : nullptr;
-    elem_t* rv = running_var.has_value() ? (elem_t*)running_var->data_ptr() : nullptr;
-    bn_finalize_kernel<elem_t><<<(C + 255) / 256, 256, 0, cur_stream()>>>(
-        sum.data_ptr<float>(), sumsq.data_ptr<float>(), C, R, (float)eps,
-        (float)momentum, mean.data_ptr<float>(), invstd.data_ptr<float>(), rm, rv);
-  });
   Tensor y = at::empty_like(x);
   int blocks = grid_for(R * C / 4, BLOCK);
   size_t lds = 2 * C * sizeof(float);
+  float* sm = saved.data_ptr<float>();
   DISPATCH_FT(x, {
-    bn_norm_kernel<elem_t><<<blocks, BLOCK, lds, cur_stream()>>>(
+    elem_t* rm = running_mean.has_value() ? (elem_t*)running_mean->data_ptr() : nullptr;
+    elem_t* rv = running_var.has_value() ? (elem_t*)running_var->data_ptr() : nullptr;
+    bn_norm_train_kernel<elem_t><<<blocks, BLOCK, lds, cur_stream()>>>(
         (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
-        mean.data_ptr<float>(), invstd.data_ptr<float>(),
+        sum.data_ptr<float>(), sumsq.data_ptr<float>(), (float)eps, (float)momentum,
         w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-        b.has_value() ? (const elem_t*)b->data_ptr() : nullptr);
+        b.has_value() ? (const elem_t*)b->data_ptr() : nullptr,
+        sm, sm + C, rm, rv);
   });
-  return {y, mean, invstd};
+  return {y, saved[0], saved[1]};
 }
 
 Tensor bn_fwd_eval(Tensor x, c10::optional<Tensor> w, c10::optional<Tensor> b,
@@ -1161,7 +1161,8 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
   TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast));
   int C = (int)x.size(1);
   auto fopt = x.options().dtype(at::kFloat);
-  Tensor sum_dy = at::zeros({C}, fopt), sum_dyx = at::zeros({C}, fopt);
+  Tensor ws = at::zeros({2, C}, fopt);
+  Tensor sum_dy = ws[0], sum_dyx = ws[1];
   hipStream_t st = cur_stream();
   DISPATCH_FT(x, {
     constexpr int N = Pack16<elem_t>::N;
@@ -1175,6 +1176,8 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
         sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
   });
   Tensor dx = at::empty_like(dy);
+  Tensor dweight = at::empty({C}, x.options());
+  Tensor dbias = at::empty({C}, x.options());
   int blocks = grid_for(R * C / 4, BLOCK);
   size_t lds = 3 * C * sizeof(float);
   DISPATCH_FT(x, {
@@ -1183,10 +1186,10 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
         (elem_t*)dx.data_ptr(), R, C, mean.data_ptr<float>(),
         invstd.data_ptr<float>(),
         w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
+        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
+        (elem_t*)dweight.data_ptr(), (elem_t*)dbias.data_ptr());
   });
-  // dweight = sum_dyx (in input dtype), dbias = sum_dy
-  return {dx, sum_dyx.to(x.scalar_type()), sum_dy.to(x.scalar_type())};
+  return {dx, dweight, dbias};
 }
 
 }  // namespace
