@@ -29,6 +29,12 @@ from torch import Tensor
 from murmura_amd.topology.base import Topology
 
 
+def _gloo_with_cuda(t: Tensor) -> bool:
+    """gloo cannot move CUDA tensors over P2P; stage through host memory
+    (used when several FL nodes share one GPU, or in CPU-comm debug runs)."""
+    return t.is_cuda and dist.get_backend() == "gloo"
+
+
 def exchange_with_neighbors(
     own: Tensor, neighbor_ids: Sequence[int], tag_base: int = 0
 ) -> Dict[int, Tensor]:
@@ -38,25 +44,29 @@ def exchange_with_neighbors(
     in one batch_isend_irecv group."""
     if not neighbor_ids:
         return {}
-    bufs = {j: torch.empty_like(own) for j in neighbor_ids}
+    stage_host = _gloo_with_cuda(own)
+    wire = own.cpu() if stage_host else own
+    bufs = {j: torch.empty_like(wire) for j in neighbor_ids}
     ops: List[dist.P2POp] = []
     # deterministic global order: send before recv per peer, peers sorted
     for j in sorted(neighbor_ids):
-        ops.append(dist.P2POp(dist.isend, own, j))
+        ops.append(dist.P2POp(dist.isend, wire, j))
         ops.append(dist.P2POp(dist.irecv, bufs[j], j))
     reqs = dist.batch_isend_irecv(ops)
     for r in reqs:
         r.wait()
+    if stage_host:
+        return {j: b.to(own.device, non_blocking=True) for j, b in bufs.items()}
     return bufs
 
 
 def allreduce_mean(own: Tensor) -> Tensor:
     """Fully-connected FedAvg fast path: global mean of flat states in one
     RCCL all-reduce (ring over xGMI)."""
-    out = own.clone()
+    out = own.cpu().clone() if _gloo_with_cuda(own) else own.clone()
     dist.all_reduce(out, op=dist.ReduceOp.SUM)
     out.div_(dist.get_world_size())
-    return out
+    return out.to(own.device) if out.device != own.device else out
 
 
 def symmetrize_wants(want: Sequence[int], world_size: int, device) -> List[List[int]]:
@@ -66,6 +76,8 @@ def symmetrize_wants(want: Sequence[int], world_size: int, device) -> List[List[
     DMTT's collaborator sets are asymmetric; RCCL P2P requires both ends to
     post matching ops, so the union set is agreed via one tiny all-gather of
     an N-bit mask (8 bytes on an 8-GPU box)."""
+    if dist.get_backend() == "gloo":
+        device = torch.device("cpu")
     mask = torch.zeros(world_size, dtype=torch.uint8, device=device)
     for j in want:
         mask[j] = 1

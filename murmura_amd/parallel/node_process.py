@@ -43,6 +43,9 @@ from murmura_amd.utils.timing import PhaseTimer
 
 
 def _resolve_backend(config: Config) -> str:
+    env = os.environ.get("MURMURA_COMM_BACKEND")
+    if env:
+        return env
     b = config.distributed.comm_backend
     if b != "auto":
         return b
@@ -61,7 +64,8 @@ def init_distributed(config: Config, rank: int, world_size: int) -> torch.device
             world_size=world_size,
             timeout=datetime.timedelta(seconds=300),
         )
-    if backend == "nccl":
+    if backend == "nccl" or (backend == "gloo" and torch.cuda.is_available()
+                              and os.environ.get("MURMURA_GLOO_CUDA") == "1"):
         local = int(os.environ.get("LOCAL_RANK", rank % max(1, torch.cuda.device_count())))
         device = torch.device(f"cuda:{local}")
         torch.cuda.set_device(device)
@@ -169,7 +173,7 @@ class FLRoundLoop:
         ready = torch.tensor(
             [1 if train_seconds <= budget else 0], dtype=torch.uint8
         )
-        if self.device.type == "cuda":
+        if self.device.type == "cuda" and dist.get_backend() != "gloo":
             ready = ready.to(self.device)
         out = [torch.zeros_like(ready) for _ in range(self.world)]
         dist.all_gather(out, ready)
@@ -250,9 +254,12 @@ class FLRoundLoop:
 
         agg = self.node.aggregator
         own_sketch = agg.get_sketch(own)  # [S]
-        all_sk = [_t.empty_like(own_sketch) for _ in range(self.world)]
-        dist.all_gather(all_sk, own_sketch.contiguous())
-        all_sk = _t.stack(all_sk)  # [N, S]
+        wire_sk = own_sketch.contiguous()
+        if wire_sk.is_cuda and dist.get_backend() == "gloo":
+            wire_sk = wire_sk.cpu()
+        all_sk = [_t.empty_like(wire_sk) for _ in range(self.world)]
+        dist.all_gather(all_sk, wire_sk)
+        all_sk = _t.stack(all_sk).to(own.device)  # [N, S]
         if not nbr_ids:
             self.node.set_state(own)
             return
