@@ -66,7 +66,8 @@ def init_distributed(config: Config, rank: int, world_size: int) -> torch.device
         )
     if backend == "nccl" or (backend == "gloo" and torch.cuda.is_available()
                               and os.environ.get("MURMURA_GLOO_CUDA") == "1"):
-        local = int(os.environ.get("LOCAL_RANK", rank % max(1, torch.cuda.device_count())))
+        ndev = max(1, torch.cuda.device_count())
+        local = int(os.environ.get("LOCAL_RANK", rank)) % ndev
         device = torch.device(f"cuda:{local}")
         torch.cuda.set_device(device)
     else:
