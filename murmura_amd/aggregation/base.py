@@ -14,7 +14,7 @@ the round loop never forces a host sync for bookkeeping.
 from __future__ import annotations
 
 import abc
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 import torch
 from torch import Tensor

@@ -8,7 +8,7 @@ compute dtype). ``extra="forbid"`` everywhere, like the reference.
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Literal, Optional
+from typing import Any, Dict, Literal, Optional
 
 from pydantic import BaseModel, ConfigDict, Field, model_validator
 

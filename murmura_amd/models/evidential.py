@@ -8,8 +8,6 @@ pass a plain MLP.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 from torch import Tensor, nn
 import torch.nn.functional as F

@@ -32,10 +32,9 @@ import torch.distributed as dist
 from torch.utils.data import DataLoader
 
 from murmura_amd.config.schema import Config
-from murmura_amd.core.network import HISTORY_KEYS, new_history
+from murmura_amd.core.network import new_history
 from murmura_amd.core.node import Node
 from murmura_amd.parallel import exchange
-from murmura_amd.topology.dynamic import MobilityModel
 from murmura_amd.topology.generators import create_topology
 from murmura_amd.utils import factories
 from murmura_amd.utils.seed import set_seed

@@ -14,7 +14,7 @@ Format (torch.save):
 from __future__ import annotations
 
 from pathlib import Path
-from typing import Dict, Optional
+from typing import Optional
 
 import torch
 

@@ -11,7 +11,7 @@ from __future__ import annotations
 import os
 import time
 from collections import defaultdict
-from typing import Dict, List
+from typing import Dict
 
 import torch
 
