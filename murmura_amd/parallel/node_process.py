@@ -336,9 +336,16 @@ def _append_history(history: Dict[str, List[float]], round_num: int, rows: List[
 
 
 def run_node_process(
-    config: Config, rank: int, world_size: int, destroy_group: bool = True
+    config: Config, rank: int, world_size: int, destroy_group: bool = True,
+    checkpoint_dir: Optional[str] = None, checkpoint_every: int = 0,
+    resume: bool = False,
 ) -> Optional[Dict[str, List[float]]]:
-    """Full distributed run for one rank; rank 0 returns the history."""
+    """Full distributed run for one rank; rank 0 returns the history.
+
+    ``checkpoint_dir``: each rank writes node_<rank>.ckpt every
+    ``checkpoint_every`` rounds; ``resume`` restores from those files."""
+    from murmura_amd.utils import checkpoint as ckpt
+
     device = init_distributed(config, rank, world_size)
     if config.dmtt is not None:
         from murmura_amd.dmtt.node_process import DMTTRoundLoop
@@ -347,7 +354,18 @@ def run_node_process(
     else:
         loop = FLRoundLoop(config, rank, world_size, device)
     history = new_history() if rank == 0 else None
-    for r in range(config.experiment.rounds):
+    start_round = 0
+    ckpt_path = None
+    if checkpoint_dir is not None:
+        import pathlib
+
+        ckpt_path = pathlib.Path(checkpoint_dir) / f"node_{rank}.ckpt"
+        if resume and ckpt_path.exists():
+            blob = ckpt.load_checkpoint(ckpt_path)
+            start_round = ckpt.restore_rank(loop.node, blob)
+            if rank == 0 and blob.get("history"):
+                history = blob["history"]
+    for r in range(start_round, config.experiment.rounds):
         loop.run_round(r)
         metrics = loop.evaluate_round(r)
         rows = exchange.gather_metrics(metrics, dst=0)
@@ -359,6 +377,9 @@ def run_node_process(
                     f"honest={history['honest_accuracy'][-1]:.4f}",
                     flush=True,
                 )
+        if ckpt_path is not None and checkpoint_every and (r + 1) % checkpoint_every == 0:
+            ckpt.save_rank_checkpoint(ckpt_path, r, loop.node,
+                                      history if rank == 0 else None)
     # aggregate per-node aggregator statistics on rank 0 (the distributed
     # analogue of Network.get_node_statistics)
     stats = loop.node.aggregator.get_statistics()

@@ -65,3 +65,15 @@ def restore_network(network, blob: dict) -> int:
     if blob.get("history"):
         network.history = blob["history"]
     return int(blob["round"]) + 1
+
+
+def save_rank_checkpoint(path, round_num: int, node, history=None) -> None:
+    """Per-rank checkpoint for the distributed backend (each rank persists its
+    own node; rank 0 additionally persists the history)."""
+    save_checkpoint(path, round_num, [node], history)
+
+
+def restore_rank(node, blob: dict) -> int:
+    if node.node_id in blob["nodes"]:
+        restore_node_state(node, blob["nodes"][node.node_id])
+    return int(blob["round"]) + 1

@@ -150,3 +150,45 @@ def test_round_budget_straggler_semantics(monkeypatch):
     h0 = _run_distributed(base, 2, 29632)
     for a, b in zip(h2["mean_accuracy"], h0["mean_accuracy"]):
         assert a == pytest.approx(b, abs=2e-3)
+
+
+def test_distributed_checkpoint_resume(tmp_path):
+    import functools
+
+    cfg = _base_config(2, algo="fedavg", topo="ring", rounds=3)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+
+    def run(extra_kwargs, port):
+        procs = [
+            ctx.Process(target=_ckpt_worker,
+                        args=(r, json.dumps(cfg), 2, port, q, str(tmp_path), extra_kwargs))
+            for r in range(2)
+        ]
+        for p in procs:
+            p.start()
+        h = q.get(timeout=240)
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+        return h
+
+    h1 = run({"checkpoint_every": 1}, 29640)
+    assert (tmp_path / "node_0.ckpt").exists()
+    assert (tmp_path / "node_1.ckpt").exists()
+    # resume: all rounds already done -> returns restored history untouched
+    h2 = run({"checkpoint_every": 1, "resume": True}, 29641)
+    assert h2["round"] == h1["round"]
+
+
+def _ckpt_worker(rank, cfg_json, world, port, q, ckpt_dir, extra):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ.pop("WORLD_SIZE", None)
+    cfg = Config(**json.loads(cfg_json))
+    cfg.distributed.master_port = port
+    from murmura_amd.parallel.node_process import run_node_process
+
+    h = run_node_process(cfg, rank, world, checkpoint_dir=ckpt_dir, **extra)
+    if rank == 0:
+        q.put(h)
