@@ -208,3 +208,32 @@ def test_wearable_config_trains_end_to_end(tmp_path):
     h = _run_simulation(cfg, verbose=False)
     assert len(h["round"]) == 2
     assert h["mean_vacuity"][-1] > 0  # evidential metrics flowing
+
+
+def test_leaf_celeba_with_images(tmp_path):
+    """CelebA loader: JSON shards reference image files; images load + resize."""
+    from PIL import Image
+
+    from murmura_amd.examples.leaf.datasets import LEAFCelebADataset
+
+    img_dir = tmp_path / "imgs"
+    img_dir.mkdir()
+    rng = np.random.default_rng(4)
+    names = []
+    for i in range(6):
+        name = f"img_{i:03d}.jpg"
+        arr = (rng.random((50, 40, 3)) * 255).astype(np.uint8)
+        Image.fromarray(arr).save(img_dir / name)
+        names.append(name)
+    d = tmp_path / "train"
+    d.mkdir()
+    blob = {"users": ["u0", "u1"], "num_samples": [3, 3],
+            "user_data": {"u0": {"x": names[:3], "y": [0, 1, 0]},
+                          "u1": {"x": names[3:], "y": [1, 1, 0]}}}
+    (d / "all_data_0.json").write_text(json.dumps(blob))
+    ds = LEAFCelebADataset(str(tmp_path), split="train", images_dir=str(img_dir))
+    assert len(ds) == 6
+    x, y = ds[0]
+    assert x.shape == (3, 84, 84)
+    assert 0.0 <= x.min() and x.max() <= 1.0
+    assert y in (0, 1)
