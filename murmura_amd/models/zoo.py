@@ -13,7 +13,7 @@ from typing import Callable, Dict
 import torch
 from torch import Tensor, nn
 
-from murmura_amd.ops.fused_bn import MurmuraBatchNorm2d
+from murmura_amd.ops.fused_bn import MurmuraBatchNorm2d, MurmuraBNReLU
 
 
 class SimpleMLP(nn.Module):
@@ -107,7 +107,7 @@ class BasicBlock(nn.Module):
     def __init__(self, in_ch: int, out_ch: int, stride: int = 1):
         super().__init__()
         self.conv1 = nn.Conv2d(in_ch, out_ch, 3, stride=stride, padding=1, bias=False)
-        self.bn1 = MurmuraBatchNorm2d(out_ch)
+        self.bn1 = MurmuraBNReLU(out_ch)  # ReLU folded into the BN kernels
         self.conv2 = nn.Conv2d(out_ch, out_ch, 3, stride=1, padding=1, bias=False)
         self.bn2 = MurmuraBatchNorm2d(out_ch)
         self.shortcut = nn.Sequential()
@@ -118,7 +118,7 @@ class BasicBlock(nn.Module):
             )
 
     def forward(self, x: Tensor) -> Tensor:
-        out = torch.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))  # BN+ReLU fused
         out = self.bn2(self.conv2(out))
         return torch.relu(out + self.shortcut(x))
 
@@ -130,7 +130,7 @@ class ResNet18(nn.Module):
     def __init__(self, num_classes: int = 10, in_channels: int = 3):
         super().__init__()
         self.conv1 = nn.Conv2d(in_channels, 64, 3, stride=1, padding=1, bias=False)
-        self.bn1 = MurmuraBatchNorm2d(64)
+        self.bn1 = MurmuraBNReLU(64)  # ReLU folded into the BN kernels
         layers = []
         in_ch = 64
         for out_ch, stride in [(64, 1), (64, 1), (128, 2), (128, 1),
@@ -141,7 +141,7 @@ class ResNet18(nn.Module):
         self.fc = nn.Linear(512, num_classes)
 
     def forward(self, x: Tensor) -> Tensor:
-        out = torch.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))  # BN+ReLU fused
         out = self.layers(out)
         out = torch.nn.functional.adaptive_avg_pool2d(out, 1).flatten(1)
         return self.fc(out)

@@ -32,22 +32,36 @@ def _supported_c(c: int) -> bool:
 
 class _FusedBNTrain(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, running_mean, running_var, momentum, eps):
+    def forward(ctx, x, weight, bias, running_mean, running_var, momentum, eps, relu):
         y, mean, invstd = _ext().bn_fwd_train(
-            x, weight, bias, running_mean, running_var, momentum, eps
+            x, weight, bias, running_mean, running_var, momentum, eps, relu
         )
-        ctx.save_for_backward(x, weight, mean, invstd)
+        ctx.relu = relu
+        if relu:
+            ctx.save_for_backward(x, weight, mean, invstd, y)
+        else:
+            ctx.save_for_backward(x, weight, mean, invstd)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, weight, mean, invstd = ctx.saved_tensors
+        if ctx.relu:
+            x, weight, mean, invstd, y = ctx.saved_tensors
+        else:
+            x, weight, mean, invstd = ctx.saved_tensors
+            y = None
         dy = dy.contiguous(memory_format=torch.channels_last)
-        dx, dweight, dbias = _ext().bn_bwd(x, dy, weight, mean, invstd)
-        return dx, dweight, dbias, None, None, None, None
+        dx, dweight, dbias = _ext().bn_bwd(x, dy, weight, mean, invstd, y, ctx.relu)
+        return dx, dweight, dbias, None, None, None, None, None
 
 
 class MurmuraBatchNorm2d(nn.BatchNorm2d):
+    """Fused NHWC BatchNorm; set ``fuse_relu=True`` (or use MurmuraBNReLU) to
+    fold the subsequent ReLU into the normalize pass and its mask into the
+    backward (one fewer elementwise kernel in each direction)."""
+
+    fuse_relu: bool = False
+
     def forward(self, x: Tensor) -> Tensor:
         use_fused = (
             x.is_cuda
@@ -59,14 +73,21 @@ class MurmuraBatchNorm2d(nn.BatchNorm2d):
             and _ext() is not None
         )
         if not use_fused:
-            return super().forward(x)
+            out = super().forward(x)
+            return torch.relu(out) if self.fuse_relu else out
         if self.training:
             if self.num_batches_tracked is not None:
                 self.num_batches_tracked.add_(1)
             return _FusedBNTrain.apply(
                 x, self.weight, self.bias, self.running_mean, self.running_var,
                 self.momentum if self.momentum is not None else 0.1, self.eps,
+                self.fuse_relu,
             )
         return _ext().bn_fwd_eval(
-            x, self.weight, self.bias, self.running_mean, self.running_var, self.eps
+            x, self.weight, self.bias, self.running_mean, self.running_var, self.eps,
+            self.fuse_relu,
         )
+
+
+class MurmuraBNReLU(MurmuraBatchNorm2d):
+    fuse_relu = True

@@ -627,7 +627,7 @@ __global__ void bn_sums_kernel(const T* __restrict__ x, int64_t R, int C,
 // save_invstd for backward and EMA-updates the running stats. Folding the
 // old finalize kernel in here saves one launch per BN call (launch count is
 // what dominates small-batch BN inside graphs).
-template <typename T>
+template <typename T, bool RELU>
 __global__ void bn_norm_train_kernel(const T* __restrict__ x, T* __restrict__ y,
                                      int64_t R, int C,
                                      const float* __restrict__ sum,
@@ -667,14 +667,17 @@ __global__ void bn_norm_train_kernel(const T* __restrict__ x, T* __restrict__ y,
     Pack16<T> pv = reinterpret_cast<const Pack16<T>*>(x)[v];
     const int cbase = (int)(v % cvec) * N;
 #pragma unroll
-    for (int k = 0; k < N; ++k)
-      from_f(pv.e[k], fmaf(to_f(pv.e[k]), scale[cbase + k], shift[cbase + k]));
+    for (int k = 0; k < N; ++k) {
+      float val = fmaf(to_f(pv.e[k]), scale[cbase + k], shift[cbase + k]);
+      if (RELU) val = fmaxf(val, 0.0f);
+      from_f(pv.e[k], val);
+    }
     reinterpret_cast<Pack16<T>*>(y)[v] = pv;
   }
 }
 
 // eval-mode forward: same as bn_norm but coefficients from running stats
-template <typename T>
+template <typename T, bool RELU>
 __global__ void bn_eval_kernel(const T* __restrict__ x, T* __restrict__ y, int64_t R,
                                int C, const T* __restrict__ running_mean,
                                const T* __restrict__ running_var, float eps,
@@ -697,16 +700,20 @@ __global__ void bn_eval_kernel(const T* __restrict__ x, T* __restrict__ y, int64
     Pack16<T> pv = reinterpret_cast<const Pack16<T>*>(x)[v];
     const int cbase = (int)(v % cvec) * N;
 #pragma unroll
-    for (int k = 0; k < N; ++k)
-      from_f(pv.e[k], fmaf(to_f(pv.e[k]), scale[cbase + k], shift[cbase + k]));
+    for (int k = 0; k < N; ++k) {
+      float val = fmaf(to_f(pv.e[k]), scale[cbase + k], shift[cbase + k]);
+      if (RELU) val = fmaxf(val, 0.0f);
+      from_f(pv.e[k], val);
+    }
     reinterpret_cast<Pack16<T>*>(y)[v] = pv;
   }
 }
 
 // backward pass 1: per-channel sum(dy) and sum(dy * xhat), same packed
 // streaming structure as bn_sums_kernel.
-template <typename T>
+template <typename T, bool RELU>
 __global__ void bn_bwd_sums_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                                   const T* __restrict__ yout,
                                    int64_t R, int C, const float* __restrict__ mean,
                                    const float* __restrict__ invstd,
                                    float* __restrict__ sum_dy,
@@ -731,9 +738,12 @@ __global__ void bn_bwd_sums_kernel(const T* __restrict__ x, const T* __restrict_
          r += (int64_t)gridDim.x * rows_per_iter) {
       Pack16<T> xv = *reinterpret_cast<const Pack16<T>*>(x + r * C + cbase);
       Pack16<T> gv = *reinterpret_cast<const Pack16<T>*>(dy + r * C + cbase);
+      Pack16<T> yv;
+      if (RELU) yv = *reinterpret_cast<const Pack16<T>*>(yout + r * C + cbase);
 #pragma unroll
       for (int k = 0; k < N; ++k) {
         float g = to_f(gv.e[k]);
+        if (RELU && to_f(yv.e[k]) <= 0.0f) g = 0.0f;
         float xh = (to_f(xv.e[k]) - m[k]) * is[k];
         sacc[k] += g;
         qacc[k] = fmaf(g, xh, qacc[k]);
@@ -758,8 +768,9 @@ __global__ void bn_bwd_sums_kernel(const T* __restrict__ x, const T* __restrict_
 }
 
 // backward pass 2: dx = w*invstd * (dy - sum_dy/R - xhat * sum_dyx/R)
-template <typename T>
+template <typename T, bool RELU>
 __global__ void bn_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                                 const T* __restrict__ yout,
                                  T* __restrict__ dx, int64_t R, int C,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ invstd,
@@ -788,12 +799,16 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ 
   for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
     Pack16<T> xv = reinterpret_cast<const Pack16<T>*>(x)[v];
     Pack16<T> gv = reinterpret_cast<const Pack16<T>*>(dy)[v];
+    Pack16<T> yv;
+    if (RELU) yv = reinterpret_cast<const Pack16<T>*>(yout)[v];
     const int cbase = (int)(v % cvec) * N;
 #pragma unroll
     for (int k = 0; k < N; ++k) {
       const int c = cbase + k;
+      float g = to_f(gv.e[k]);
+      if (RELU && to_f(yv.e[k]) <= 0.0f) g = 0.0f;
       float xh = (to_f(xv.e[k]) - mean[c]) * invstd[c];
-      float t = to_f(gv.e[k]) - g_mean[c] - xh * g_proj[c];
+      float t = g - g_mean[c] - xh * g_proj[c];
       from_f(gv.e[k], t * g_scale[c]);
     }
     reinterpret_cast<Pack16<T>*>(dx)[v] = gv;
@@ -1110,7 +1125,7 @@ void bn_sums_dispatch(const Tensor& x, int64_t R, int C, Tensor& sum, Tensor& su
 std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optional<Tensor> b,
                                  c10::optional<Tensor> running_mean,
                                  c10::optional<Tensor> running_var, double momentum,
-                                 double eps) {
+                                 double eps, bool relu) {
   int64_t R = bn_check(x);
   int C = (int)x.size(1);
   auto fopt = x.options().dtype(at::kFloat);
@@ -1129,18 +1144,27 @@ std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optiona
   DISPATCH_FT(x, {
     elem_t* rm = running_mean.has_value() ? (elem_t*)running_mean->data_ptr() : nullptr;
     elem_t* rv = running_var.has_value() ? (elem_t*)running_var->data_ptr() : nullptr;
-    bn_norm_train_kernel<elem_t><<<blocks, BLOCK, lds, cur_stream()>>>(
-        (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
-        sum.data_ptr<float>(), sumsq.data_ptr<float>(), (float)eps, (float)momentum,
-        w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-        b.has_value() ? (const elem_t*)b->data_ptr() : nullptr,
-        sm, sm + C, rm, rv);
+    if (relu) {
+      bn_norm_train_kernel<elem_t, true><<<blocks, BLOCK, lds, cur_stream()>>>(
+          (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
+          sum.data_ptr<float>(), sumsq.data_ptr<float>(), (float)eps, (float)momentum,
+          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr,
+          sm, sm + C, rm, rv);
+    } else {
+      bn_norm_train_kernel<elem_t, false><<<blocks, BLOCK, lds, cur_stream()>>>(
+          (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
+          sum.data_ptr<float>(), sumsq.data_ptr<float>(), (float)eps, (float)momentum,
+          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr,
+          sm, sm + C, rm, rv);
+    }
   });
   return {y, saved[0], saved[1]};
 }
 
 Tensor bn_fwd_eval(Tensor x, c10::optional<Tensor> w, c10::optional<Tensor> b,
-                   Tensor running_mean, Tensor running_var, double eps) {
+                   Tensor running_mean, Tensor running_var, double eps, bool relu) {
   int64_t R = bn_check(x);
   int C = (int)x.size(1);
   Tensor y = at::empty_like(x);
@@ -1149,18 +1173,28 @@ Tensor bn_fwd_eval(Tensor x, c10::optional<Tensor> w, c10::optional<Tensor> b,
   TORCH_CHECK(running_mean.scalar_type() == x.scalar_type(),
               "bn: running stats must match input dtype");
   DISPATCH_FT(x, {
-    bn_eval_kernel<elem_t><<<blocks, BLOCK, lds, cur_stream()>>>(
-        (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
-        (const elem_t*)running_mean.data_ptr(),
-        (const elem_t*)running_var.data_ptr(), (float)eps,
-        w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-        b.has_value() ? (const elem_t*)b->data_ptr() : nullptr);
+    if (relu) {
+      bn_eval_kernel<elem_t, true><<<blocks, BLOCK, lds, cur_stream()>>>(
+          (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
+          (const elem_t*)running_mean.data_ptr(),
+          (const elem_t*)running_var.data_ptr(), (float)eps,
+          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr);
+    } else {
+      bn_eval_kernel<elem_t, false><<<blocks, BLOCK, lds, cur_stream()>>>(
+          (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
+          (const elem_t*)running_mean.data_ptr(),
+          (const elem_t*)running_var.data_ptr(), (float)eps,
+          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr);
+    }
   });
   return y;
 }
 
 std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor mean,
-                           Tensor invstd) {
+                           Tensor invstd, c10::optional<Tensor> yout, bool relu) {
+  TORCH_CHECK(!relu || yout.has_value(), "bn_bwd: relu needs the saved output");
   int64_t R = bn_check(x);
   TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast));
   int C = (int)x.size(1);
@@ -1174,10 +1208,18 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
     const int rows_per_iter = std::max(1, BLOCK / packs_per_row);
     int blocks = grid_for(R, rows_per_iter, 256);  // see bn_sums_dispatch
     size_t lds = 2 * C * sizeof(float);
-    bn_bwd_sums_kernel<elem_t><<<blocks, BLOCK, lds, st>>>(
-        (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), R, C,
-        mean.data_ptr<float>(), invstd.data_ptr<float>(),
-        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
+    const elem_t* yp = yout.has_value() ? (const elem_t*)yout->data_ptr() : nullptr;
+    if (relu) {
+      bn_bwd_sums_kernel<elem_t, true><<<blocks, BLOCK, lds, st>>>(
+          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp, R, C,
+          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
+    } else {
+      bn_bwd_sums_kernel<elem_t, false><<<blocks, BLOCK, lds, st>>>(
+          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp, R, C,
+          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
+    }
   });
   Tensor dx = at::empty_like(dy);
   Tensor dweight = at::empty({C}, x.options());
@@ -1185,13 +1227,24 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
   int blocks = grid_for(R * C / 4, BLOCK);
   size_t lds = 3 * C * sizeof(float);
   DISPATCH_FT(x, {
-    bn_bwd_dx_kernel<elem_t><<<blocks, BLOCK, lds, st>>>(
-        (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(),
-        (elem_t*)dx.data_ptr(), R, C, mean.data_ptr<float>(),
-        invstd.data_ptr<float>(),
-        w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
-        (elem_t*)dweight.data_ptr(), (elem_t*)dbias.data_ptr());
+    const elem_t* yp2 = yout.has_value() ? (const elem_t*)yout->data_ptr() : nullptr;
+    if (relu) {
+      bn_bwd_dx_kernel<elem_t, true><<<blocks, BLOCK, lds, st>>>(
+          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp2,
+          (elem_t*)dx.data_ptr(), R, C, mean.data_ptr<float>(),
+          invstd.data_ptr<float>(),
+          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
+          (elem_t*)dweight.data_ptr(), (elem_t*)dbias.data_ptr());
+    } else {
+      bn_bwd_dx_kernel<elem_t, false><<<blocks, BLOCK, lds, st>>>(
+          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp2,
+          (elem_t*)dx.data_ptr(), R, C, mean.data_ptr<float>(),
+          invstd.data_ptr<float>(),
+          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
+          (elem_t*)dweight.data_ptr(), (elem_t*)dbias.data_ptr());
+    }
   });
   return {dx, dweight, dbias};
 }
@@ -1200,9 +1253,16 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train,
-        "K14: fused NHWC BatchNorm training forward -> (y, mean, invstd)");
-  m.def("bn_fwd_eval", &bn_fwd_eval, "K14: NHWC BatchNorm eval forward");
-  m.def("bn_bwd", &bn_bwd, "K14: NHWC BatchNorm backward -> (dx, dweight, dbias)");
+        "K14: fused NHWC BatchNorm(+ReLU) training forward -> (y, mean, invstd)",
+        py::arg("x"), py::arg("w"), py::arg("b"), py::arg("running_mean"),
+        py::arg("running_var"), py::arg("momentum"), py::arg("eps"),
+        py::arg("relu") = false);
+  m.def("bn_fwd_eval", &bn_fwd_eval, "K14: NHWC BatchNorm(+ReLU) eval forward",
+        py::arg("x"), py::arg("w"), py::arg("b"), py::arg("running_mean"),
+        py::arg("running_var"), py::arg("eps"), py::arg("relu") = false);
+  m.def("bn_bwd", &bn_bwd, "K14: NHWC BatchNorm(+ReLU) backward -> (dx, dweight, dbias)",
+        py::arg("x"), py::arg("dy"), py::arg("w"), py::arg("mean"), py::arg("invstd"),
+        py::arg("yout") = py::none(), py::arg("relu") = false);
   m.def("weighted_sum", &weighted_sum, "K1: out = sum_i w_i * x_i",
         py::arg("stacked"), py::arg("w"), py::arg("out") = py::none());
   m.def("pairwise_sq_dists", &pairwise_sq_dists, "K2: [m,m] squared L2 matrix");

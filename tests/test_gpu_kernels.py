@@ -251,3 +251,35 @@ def test_fused_bn_cpu_fallback_identical_to_torch():
     r.load_state_dict(m.state_dict())
     x = torch.randn(4, 16, 5, 5)
     assert torch.allclose(m(x), r(x))
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_bn_relu_matches_torch(dtype):
+    from murmura_amd.ops.fused_bn import MurmuraBNReLU
+
+    torch.manual_seed(5)
+    c = 128
+    x = torch.randn(8, c, 8, 8, device="cuda", dtype=dtype)
+    xf = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    xr = x.clone().float().requires_grad_(True)
+    fused = MurmuraBNReLU(c).cuda().to(dtype)
+    refbn = nn.BatchNorm2d(c).cuda().float()
+    refbn.load_state_dict({k: v.float() for k, v in fused.state_dict().items()})
+    fused.train(); refbn.train()
+    y = fused(xf)
+    yr = torch.relu(refbn(xr))
+    tol = 1e-5 if dtype == torch.float32 else 3e-2
+    assert torch.all(y >= 0)
+    assert torch.allclose(y.float(), yr, atol=tol, rtol=tol)
+    g = torch.randn_like(yr)
+    y.backward(g.to(dtype).contiguous(memory_format=torch.channels_last))
+    yr.backward(g)
+    gtol = 1e-4 if dtype == torch.float32 else 6e-2
+    scale = xr.grad.abs().max().clamp_min(1e-6)
+    assert ((xf.grad.float() - xr.grad).abs().max() / scale).item() < gtol
+    # eval-mode fused relu
+    fused.eval(); refbn.eval()
+    with torch.no_grad():
+        ye = fused(x.contiguous(memory_format=torch.channels_last))
+        yre = torch.relu(refbn(x.float()))
+    assert torch.allclose(ye.float(), yre, atol=tol, rtol=tol)
