@@ -83,6 +83,8 @@ class TrainGraph:
         self.static_y = torch.zeros((n_used,), device=shard.device, dtype=shard.y.dtype)
         self.loss_sum = torch.zeros((), device=shard.device, dtype=torch.float32)
         self.lr = torch.zeros((), device=shard.device, dtype=torch.float32)
+        # device-resident KL annealing weight for EvidentialLoss capture
+        self.kl_weight = torch.zeros((), device=shard.device, dtype=torch.float32)
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self._gen = torch.Generator().manual_seed(
             0x5EED ^ (node.node_id * 0x9E3779B9 & 0x7FFFFFFF)
@@ -92,9 +94,14 @@ class TrainGraph:
         node = self.node
         x = self.static_x[s * self.bs : (s + 1) * self.bs]
         y = self.static_y[s * self.bs : (s + 1) * self.bs]
+        from murmura_amd.models.evidential import EvidentialLoss
+
         node.store.zero_grad()
         out = node.model(x)
-        loss = torch.nn.functional.cross_entropy(out.float(), y)
+        if isinstance(node.criterion, EvidentialLoss):
+            loss = node.criterion(out, y, kl_weight=self.kl_weight)
+        else:
+            loss = torch.nn.functional.cross_entropy(out.float(), y)
         loss.backward()
         # fused SGD with a DEVICE-tensor lr so replays honor lr changes
         from murmura_amd import ops as _ops
@@ -126,12 +133,16 @@ class TrainGraph:
                 self._one_step(step)
         self.graph = g
 
-    def run_epoch(self, lr: float) -> Tensor:
+    def run_epoch(self, lr: float, round_num: int = 0) -> Tensor:
         """Shuffle + replay one epoch; returns the summed loss (device scalar)."""
+        from murmura_amd.models.evidential import EvidentialLoss
+
         # fill static buffers BEFORE the first capture: warmup executes real
         # kernels and must see genuine inputs/targets
         self.shard.shuffled(self.static_x, self.static_y, self._gen)
         self.lr.fill_(lr)
+        if isinstance(self.node.criterion, EvidentialLoss):
+            self.kl_weight.fill_(self.node.criterion.kl_weight_at(round_num))
         if self.graph is None:
             self._capture()
         self.graph.replay()

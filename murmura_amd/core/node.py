@@ -100,9 +100,8 @@ class Node:
         replay over the device-resident shard (core/gpu_round.py); the eager
         path below is the CPU oracle and the evidential/odd-shard fallback.
         """
-        if (
-            self._graphs_enabled()
-            and not isinstance(self.criterion, EvidentialLoss)
+        if self._graphs_enabled() and isinstance(
+            self.criterion, (nn.CrossEntropyLoss, EvidentialLoss)
         ):
             shard = self._get_shard("train")
             bs = getattr(self.train_loader, "batch_size", None) or 32
@@ -113,7 +112,7 @@ class Node:
                     self._train_graph = TrainGraph(self, shard, bs)
                 total = torch.zeros((), device=self.device)
                 for _ in range(max(1, epochs)):
-                    total = total + self._train_graph.run_epoch(lr)
+                    total = total + self._train_graph.run_epoch(lr, round_num)
                 nb = self._train_graph.num_batches * max(1, epochs)
                 return {"loss": float(total.item() / nb), "num_batches": nb}
         self.model.train()

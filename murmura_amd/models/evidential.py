@@ -55,7 +55,17 @@ class EvidentialLoss(nn.Module):
         self.annealing_rounds = max(1, int(annealing_rounds))
         self.max_kl_weight = float(max_kl_weight)
 
-    def forward(self, logits: Tensor, targets: Tensor, round_num: int = 0) -> Tensor:
+    def kl_weight_at(self, round_num: int) -> float:
+        return self.max_kl_weight * min(1.0, round_num / self.annealing_rounds)
+
+    def forward(
+        self, logits: Tensor, targets: Tensor, round_num: int = 0,
+        kl_weight: Tensor = None,
+    ) -> Tensor:
+        """``kl_weight``: optional 0-dim DEVICE tensor overriding the
+        round-derived annealing weight — this is what makes the loss
+        hipGraph-capturable (the graph replays with the tensor updated
+        per round instead of baking in a Python float)."""
         logits = logits.float()
         alpha = F.softplus(logits) + 1.0
         s = alpha.sum(dim=1, keepdim=True)
@@ -66,7 +76,7 @@ class EvidentialLoss(nn.Module):
         # KL(Dir(alpha_tilde) || Dir(1)) on the misleading evidence
         alpha_t = y + (1.0 - y) * alpha
         kl = self._kl_to_uniform(alpha_t)
-        lam = self.max_kl_weight * min(1.0, round_num / self.annealing_rounds)
+        lam = kl_weight if kl_weight is not None else self.kl_weight_at(round_num)
         return (mse + lam * kl).mean()
 
     def _kl_to_uniform(self, alpha: Tensor) -> Tensor:

@@ -106,3 +106,36 @@ def test_evidential_eval_graph():
     for k in ["vacuity", "entropy", "strength", "accuracy", "loss"]:
         assert torch.isfinite(res[k]), k
     assert 0.0 < res["vacuity"].item() <= 1.0
+
+
+def test_evidential_train_graph_learns_and_anneals():
+    """Evidential models train through the captured graph: the KL weight is a
+    device tensor the replay reads, so annealing still works across rounds."""
+    from murmura_amd.aggregation import EvidentialTrustAggregator
+    from murmura_amd.core.node import Node
+    from murmura_amd.data.synthetic import make_synthetic_classification
+    from murmura_amd.models import EvidentialHARClassifier, get_evidential_loss
+
+    torch.manual_seed(11)
+    ds = make_synthetic_classification(512, num_features=561, num_classes=6, seed=7)
+    node = Node(
+        0,
+        EvidentialHARClassifier(),
+        DataLoader(ds, batch_size=64, shuffle=True,
+                   generator=torch.Generator().manual_seed(2), drop_last=True),
+        DataLoader(ds, batch_size=64),
+        EvidentialTrustAggregator(),
+        torch.device("cuda:0"),
+        criterion=get_evidential_loss(6, total_rounds=20),
+        evidential=True,
+    )
+    acc0 = node.evaluate()["accuracy"].item()
+    losses = []
+    for r in range(6):
+        stats = node.local_train(epochs=1, lr=0.01, round_num=r)
+        losses.append(stats["loss"])
+    assert node._train_graph is not None  # graph path engaged
+    acc1 = node.evaluate()["accuracy"].item()
+    assert acc1 > max(acc0, 0.5)
+    # kl weight actually annealed on-device
+    assert node._train_graph.kl_weight.item() > 0.0
