@@ -80,10 +80,14 @@ class EvidentialLoss(nn.Module):
         return (mse + lam * kl).mean()
 
     def _kl_to_uniform(self, alpha: Tensor) -> Tensor:
+        import math
+
         k = float(self.num_classes)
         s = alpha.sum(dim=1)
         ln_b = torch.lgamma(alpha).sum(dim=1) - torch.lgamma(s)
-        ln_b_uni = torch.lgamma(torch.tensor(k, device=alpha.device))  # lgamma(K)·(-1) trick: ln B(1)=  -lgamma(K)
+        # ln B(1,...,1) = -lgamma(K); a plain float keeps this
+        # hipGraph-capturable (no H2D copy mid-capture)
+        ln_b_uni = math.lgamma(k)
         dg_s = torch.digamma(s).unsqueeze(1)
         term = ((alpha - 1.0) * (torch.digamma(alpha) - dg_s)).sum(dim=1)
         return ln_b_uni - ln_b + term
