@@ -118,6 +118,10 @@ def main():
         sys.exit(2)
 
     config = build_config(args, world)
+    if args.attack != "none" and world == 1 and rank == 0:
+        print("warning: with N=1 the single node is selected as compromised "
+              "and skips training; attack configs are meaningful for N>=2",
+              file=sys.stderr)
     use_cuda = torch.cuda.is_available()
 
     import torch.distributed as dist
