@@ -59,6 +59,11 @@ struct Pack16 {
   T e[N];
 };
 
+static inline int env_int(const char* name, int dflt) {
+  const char* v = getenv(name);
+  return v ? atoi(v) : dflt;
+}
+
 static inline int grid_for(int64_t work_items, int per_block, int cap = 4096) {
   int64_t b = (work_items + per_block - 1) / per_block;
   if (b < 1) b = 1;
@@ -1116,7 +1121,8 @@ void bn_sums_dispatch(const Tensor& x, int64_t R, int C, Tensor& sum, Tensor& su
   // addresses, so 2048 blocks serialize ~2048 adds per address (measured
   // 31 us for an 8 MB reduction); 256 blocks iterate instead and contend 8x
   // less while still covering all 256 CUs
-  int blocks = grid_for(R, rows_per_iter, 256);
+  static const int cap = env_int("MURMURA_BN_GRID_CAP", 64);
+  int blocks = grid_for(R, rows_per_iter, cap);
   size_t lds = 2 * C * sizeof(float);
   bn_sums_kernel<elem_t><<<blocks, BLOCK, lds, cur_stream()>>>(
       (const elem_t*)x.data_ptr(), R, C, sum.data_ptr<float>(), sumsq.data_ptr<float>());
@@ -1206,7 +1212,8 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
     constexpr int N = Pack16<elem_t>::N;
     const int packs_per_row = C / N;
     const int rows_per_iter = std::max(1, BLOCK / packs_per_row);
-    int blocks = grid_for(R, rows_per_iter, 256);  // see bn_sums_dispatch
+    static const int cap2 = env_int("MURMURA_BN_GRID_CAP", 64);
+    int blocks = grid_for(R, rows_per_iter, cap2);
     size_t lds = 2 * C * sizeof(float);
     const elem_t* yp = yout.has_value() ? (const elem_t*)yout->data_ptr() : nullptr;
     if (relu) {
