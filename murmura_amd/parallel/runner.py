@@ -37,6 +37,9 @@ class DistributedRunner:
             )
         os.environ["MASTER_ADDR"] = self.config.distributed.master_addr
         os.environ["MASTER_PORT"] = str(self.config.distributed.master_port)
+        # avoid CPU thread oversubscription: N worker processes on one host
+        if not torch.cuda.is_available():
+            os.environ.setdefault("OMP_NUM_THREADS", "2")
         ctx = mp.get_context("spawn")
         queue = ctx.SimpleQueue()
         cfg_json = self.config.model_dump_json()
