@@ -101,12 +101,17 @@ def _run_simulation(config, verbose: bool, checkpoint=None, checkpoint_every=0,
 @app.command("run-node")
 def run_node(
     config_path: Path = typer.Argument(...),
-    node_id: int = typer.Option(..., "--node-id", "-n"),
+    node_id: Optional[int] = typer.Option(
+        None, "--node-id", "-n",
+        help="rank of this node; defaults to the RANK env var (torchrun)",
+    ),
     world_size: Optional[int] = typer.Option(None, "--world-size"),
     master_addr: Optional[str] = typer.Option(None, "--master-addr"),
     master_port: Optional[int] = typer.Option(None, "--master-port"),
 ) -> None:
-    """Launch a single FL node process (multi-machine / manual launching)."""
+    """Launch a single FL node process (multi-machine / manual / torchrun)."""
+    import os
+
     from murmura_amd.config.loader import load_config
     from murmura_amd.parallel.node_process import run_node_process
 
@@ -115,7 +120,12 @@ def run_node(
         config.distributed.master_addr = master_addr
     if master_port:
         config.distributed.master_port = master_port
-    ws = world_size or config.topology.num_nodes
+    if node_id is None:
+        env_rank = os.environ.get("RANK")
+        if env_rank is None:
+            raise typer.BadParameter("--node-id required (or set RANK)")
+        node_id = int(env_rank)
+    ws = world_size or int(os.environ.get("WORLD_SIZE", config.topology.num_nodes))
     history = run_node_process(config, rank=node_id, world_size=ws)
     if history is not None:
         _display_results(history, config)
