@@ -35,7 +35,8 @@ def run_one(path: Path, timeout_s: float) -> dict:
     else:
         history = DistributedRunner(cfg).run()
     elapsed = time.time() - t0
-    final = {k: (v[-1] if v else None) for k, v in history.items()}
+    final = {k: (v[-1] if v else None) for k, v in history.items()
+             if isinstance(v, list)}
     return {
         "name": cfg.experiment.name,
         "status": "ok",
