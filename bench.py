@@ -143,10 +143,21 @@ def main():
 
     loop = FLRoundLoop(config, rank, world, device)
 
+    pending = [None]
+
     def one_round(r):
         loop.run_round(r)
         if not args.no_eval:
-            loop.evaluate_round(r)
+            # async: this round's training overlapped the PREVIOUS round's
+            # eval; resolve it before launching the next (shared accumulator)
+            if pending[0] is not None:
+                pending[0].resolve()
+            pending[0] = loop.evaluate_round_async(r)
+
+    def drain_eval():
+        if pending[0] is not None:
+            pending[0].resolve()
+            pending[0] = None
 
     def sync():
         if use_cuda:
@@ -158,6 +169,7 @@ def main():
     # warmup
     for r in range(args.warmup):
         one_round(r)
+    drain_eval()
     sync()
     # timing means should reflect steady state, not capture/comm-init rounds
     loop.timer.totals.clear()
@@ -165,6 +177,7 @@ def main():
     t0 = time.perf_counter()
     for r in range(args.warmup, args.warmup + args.steps):
         one_round(r)
+    drain_eval()  # the last round's eval is part of the timed region
     sync()
     elapsed = time.perf_counter() - t0
 

@@ -301,6 +301,40 @@ class FLRoundLoop:
         )
         self.node.set_state(new_state)
 
+    def evaluate_round_async(self, round_num: int):
+        """Launch evaluation of the current (post-aggregation) state on a side
+        stream; returns a handle whose .resolve() yields metric tensors. The
+        next round's training overlaps with this evaluation — the evaluator
+        snapshots the state first, so there is no race. Falls back to a
+        synchronous pseudo-handle on CPU."""
+        if self.device.type == "cuda" and self.node.model_factory is not None:
+            if not hasattr(self, "_async_eval") or self._async_eval is None:
+                from murmura_amd.core.async_eval import AsyncEvaluator
+
+                try:
+                    self._async_eval = AsyncEvaluator(self.node)
+                except Exception:
+                    self._async_eval = False  # unsupported model; stay sync
+            if self._async_eval:
+                return self._async_eval.launch(round_num)
+
+        class _Sync:
+            def __init__(s2, res, rn):
+                s2._res, s2.round_num = res, rn
+
+            def resolve(s2):
+                return s2._res
+
+        return _Sync(self.node.evaluate(), round_num)
+
+    def metrics_from(self, handle) -> Dict[str, float]:
+        res = handle.resolve()
+        out = {k: float(v) for k, v in res.items() if isinstance(v, torch.Tensor)}
+        out["round"] = handle.round_num
+        out["node_id"] = self.rank
+        out["compromised"] = self._is_compromised(self.rank)
+        return out
+
     def evaluate_round(self, round_num: int) -> Dict[str, float]:
         with self.timer.phase("evaluate"):
             res = self.node.evaluate()
@@ -365,21 +399,38 @@ def run_node_process(
             start_round = ckpt.restore_rank(loop.node, blob)
             if rank == 0 and blob.get("history"):
                 history = blob["history"]
-    for r in range(start_round, config.experiment.rounds):
-        loop.run_round(r)
-        metrics = loop.evaluate_round(r)
+    def _flush(handle) -> None:
+        metrics = loop.metrics_from(handle)
         rows = exchange.gather_metrics(metrics, dst=0)
         if rank == 0:
-            _append_history(history, r, rows)
+            _append_history(history, handle.round_num, rows)
             if config.experiment.verbose:
                 print(
-                    f"[round {r}] acc={history['mean_accuracy'][-1]:.4f} "
+                    f"[round {handle.round_num}] "
+                    f"acc={history['mean_accuracy'][-1]:.4f} "
                     f"honest={history['honest_accuracy'][-1]:.4f}",
                     flush=True,
                 )
+
+    # software-pipelined loop: round r+1's training overlaps round r's
+    # side-stream evaluation + metrics gather
+    pending = None
+    for r in range(start_round, config.experiment.rounds):
+        loop.run_round(r)
+        # resolve round r-1's eval FIRST: handles share the eval graph's
+        # accumulator, so the next launch must not be enqueued before the
+        # previous result is read (it already overlapped with this round's
+        # training, which is the point)
+        if pending is not None:
+            _flush(pending)
+        pending = loop.evaluate_round_async(r)
         if ckpt_path is not None and checkpoint_every and (r + 1) % checkpoint_every == 0:
+            _flush(pending)
+            pending = None
             ckpt.save_rank_checkpoint(ckpt_path, r, loop.node,
                                       history if rank == 0 else None)
+    if pending is not None:
+        _flush(pending)
     # aggregate per-node aggregator statistics on rank 0 (the distributed
     # analogue of Network.get_node_statistics)
     stats = loop.node.aggregator.get_statistics()
