@@ -139,3 +139,58 @@ def test_evidential_train_graph_learns_and_anneals():
     assert acc1 > max(acc0, 0.5)
     # kl weight actually annealed on-device
     assert node._train_graph.kl_weight.item() > 0.0
+
+
+def test_async_eval_matches_sync_eval():
+    """Side-stream async evaluation must produce the same metrics as the
+    synchronous path on the same state."""
+    from murmura_amd.core.async_eval import AsyncEvaluator
+
+    node = _make_node(graphs=True)
+    node.local_train(epochs=1, lr=0.1)
+    sync_res = node.evaluate()
+    ev = AsyncEvaluator(node)
+    h = ev.launch(0)
+    res = h.resolve()
+    assert abs(res["accuracy"].item() - sync_res["accuracy"].item()) < 1e-3
+    assert abs(res["loss"].item() - sync_res["loss"].item()) < 1e-3
+    # mutate the live state AFTER launching: the handle must still report the
+    # snapshot's metrics (launch before mutation)
+    h2 = ev.launch(1)
+    node.set_state(torch.zeros_like(node.store.flat))
+    res2 = h2.resolve()
+    assert abs(res2["accuracy"].item() - sync_res["accuracy"].item()) < 1e-3
+
+
+def test_async_eval_pipeline_in_round_loop():
+    import torch.distributed as dist
+
+    from murmura_amd.config.schema import Config
+    from murmura_amd.parallel.node_process import FLRoundLoop
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29750")
+        dist.init_process_group(backend="nccl", rank=0, world_size=1)
+    cfg = Config(**{
+        "experiment": {"rounds": 4, "verbose": False},
+        "topology": {"type": "fully", "num_nodes": 1},
+        "aggregation": {"algorithm": "fedavg"},
+        "training": {"batch_size": 32, "lr": 0.05},
+        "data": {"adapter": "synthetic",
+                 "params": {"num_samples": 256, "num_features": 16, "num_classes": 4}},
+        "model": {"factory": "models.mlp",
+                  "params": {"in_features": 16, "hidden": 32, "num_classes": 4}},
+        "backend": "rccl",
+    })
+    loop = FLRoundLoop(cfg, 0, 1, torch.device("cuda:0"))
+    pending = None
+    metrics = []
+    for r in range(4):
+        loop.run_round(r)
+        if pending is not None:
+            metrics.append(loop.metrics_from(pending))
+        pending = loop.evaluate_round_async(r)
+    metrics.append(loop.metrics_from(pending))
+    assert [m["round"] for m in metrics] == [0, 1, 2, 3]
+    assert metrics[-1]["accuracy"] > metrics[0]["accuracy"] - 0.2  # learning-ish
