@@ -25,6 +25,7 @@ def _make_node(graphs: bool):
         DataLoader(ds, batch_size=64),
         FedAvgAggregator(),
         torch.device("cuda:0"),
+        model_factory=lambda: SimpleMLP(16, 32, 4),
     )
     return node
 
