@@ -51,13 +51,24 @@ class BALANCEAggregator(Aggregator):
         own_state: Tensor,
         neighbor_states: Tensor,
         round_num: int = 0,
+        pairwise_d2: Tensor = None,
         **ctx: Any,
     ) -> Tensor:
+        """``pairwise_d2``: optional [1+k, 1+k] squared distances over
+        [own] + neighbors (chunked-exchange path); row 0 gives the distances
+        to own and entry (0, 0)... the own norm comes via the caller's Gram
+        diagonal, so we derive both from it when present."""
         k = neighbor_states.shape[0]
         if k == 0:
             return own_state.clone()
-        own_norm = ops.row_norms(own_state.unsqueeze(0)).squeeze(0)
-        dists = ops.l2_dists_to(own_state, neighbor_states)
+        if pairwise_d2 is not None:
+            dists = pairwise_d2[0, 1:].clamp_min(0).sqrt()
+            own_norm = ctx.get("own_norm")
+            if own_norm is None:
+                own_norm = ops.row_norms(own_state.unsqueeze(0)).squeeze(0)
+        else:
+            own_norm = ops.row_norms(own_state.unsqueeze(0)).squeeze(0)
+            dists = ops.l2_dists_to(own_state, neighbor_states)
         threshold = self._decay(round_num) * own_norm
         accept = dists <= threshold
         w = accept_weights(accept, dists, self.min_neighbors)

@@ -34,14 +34,19 @@ class KrumAggregator(Aggregator):
         own_state: Tensor,
         neighbor_states: Tensor,
         round_num: int = 0,
+        pairwise_d2: Tensor = None,
         **ctx: Any,
     ) -> Tensor:
+        """``pairwise_d2``: optional precomputed [m, m] squared-distance
+        matrix over [own] + neighbors in order — supplied by the chunked
+        exchange path, which accumulates the Gram matrix while state chunks
+        are still on the xGMI wire."""
         m = 1 + neighbor_states.shape[0]
         if self.num_compromised >= (m - 2) / 2 or m < 3:
             self._fallbacks += 1
             return own_state.clone()
         stacked = torch.cat([own_state.unsqueeze(0), neighbor_states], dim=0)
-        d2 = ops.pairwise_sq_dists(stacked)
+        d2 = pairwise_d2 if pairwise_d2 is not None else ops.pairwise_sq_dists(stacked)
         idx = ops.krum_select(d2, self.num_compromised)
         self._selections += 1
         return stacked.index_select(0, idx.view(1)).squeeze(0).clone()

@@ -73,6 +73,22 @@ def pairwise_l2(stacked: Tensor) -> Tensor:
     return pairwise_sq_dists(stacked).clamp_min(0).sqrt()
 
 
+def gram(x: Tensor) -> Tensor:
+    """Full [m, m] Gram matrix; accepts chunked column views (stride(0) > n)
+    so exchange can accumulate it chunk-by-chunk while later chunks are still
+    on the wire."""
+    if _use_native(x):
+        xx = x if x.stride(1) == 1 else x.contiguous()
+        return _EXT.gram(xx)
+    xf = x.float()
+    return xf @ xf.t()
+
+
+def sq_dists_from_gram(g: Tensor) -> Tensor:
+    sq = g.diagonal()
+    return (sq.unsqueeze(0) + sq.unsqueeze(1) - 2.0 * g).clamp_min(0.0)
+
+
 def row_norms(stacked: Tensor) -> Tensor:
     if _use_native(stacked):
         return _EXT.row_norms(stacked.view(1, -1) if stacked.dim() == 1 else stacked).view(

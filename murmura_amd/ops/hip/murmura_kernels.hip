@@ -119,7 +119,8 @@ __global__ void weighted_sum_kernel(const T* __restrict__ x,
 // Gram matrix over rows: gram[i, j] = <x_i, x_j> for i <= j, each row read
 // exactly once (register-tiled over compile-time M).
 template <typename T, int M>
-__global__ void gram_kernel(const T* __restrict__ x, int64_t P, float* __restrict__ gram) {
+__global__ void gram_kernel(const T* __restrict__ x, int64_t P, int64_t ld,
+                            float* __restrict__ gram) {
   constexpr int NPAIR = M * (M + 1) / 2;
   constexpr int N = Pack16<T>::N;
   __shared__ float lacc[NPAIR];
@@ -135,7 +136,7 @@ __global__ void gram_kernel(const T* __restrict__ x, int64_t P, float* __restric
   for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
     Pack16<T> rows[M];
 #pragma unroll
-    for (int i = 0; i < M; ++i) rows[i] = reinterpret_cast<const Pack16<T>*>(x + (int64_t)i * P)[v];
+    for (int i = 0; i < M; ++i) rows[i] = reinterpret_cast<const Pack16<T>*>(x + (int64_t)i * ld)[v];
     int t = 0;
 #pragma unroll
     for (int i = 0; i < M; ++i) {
@@ -153,7 +154,7 @@ __global__ void gram_kernel(const T* __restrict__ x, int64_t P, float* __restric
   for (int64_t p = nvec * N + blockIdx.x * blockDim.x + threadIdx.x; p < P; p += stride) {
     float rowsf[M];
 #pragma unroll
-    for (int i = 0; i < M; ++i) rowsf[i] = to_f(x[(int64_t)i * P + p]);
+    for (int i = 0; i < M; ++i) rowsf[i] = to_f(x[(int64_t)i * ld + p]);
     int t = 0;
 #pragma unroll
     for (int i = 0; i < M; ++i)
@@ -178,7 +179,8 @@ __global__ void gram_kernel(const T* __restrict__ x, int64_t P, float* __restric
 // packs in registers at the cost of re-reading rows once per tile pair.
 template <typename T, int MI, int MJ>
 __global__ void gram_cross_kernel(const T* __restrict__ xi, const T* __restrict__ xj,
-                                  int64_t P, float* __restrict__ out, int ldo) {
+                                  int64_t P, int64_t ld, float* __restrict__ out,
+                                  int ldo) {
   constexpr int N = Pack16<T>::N;
   __shared__ float lacc[MI * MJ];
   for (int t = threadIdx.x; t < MI * MJ; t += blockDim.x) lacc[t] = 0.0f;
@@ -193,9 +195,9 @@ __global__ void gram_cross_kernel(const T* __restrict__ xi, const T* __restrict_
   for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
     Pack16<T> ri[MI], rj[MJ];
 #pragma unroll
-    for (int i = 0; i < MI; ++i) ri[i] = reinterpret_cast<const Pack16<T>*>(xi + (int64_t)i * P)[v];
+    for (int i = 0; i < MI; ++i) ri[i] = reinterpret_cast<const Pack16<T>*>(xi + (int64_t)i * ld)[v];
 #pragma unroll
-    for (int j = 0; j < MJ; ++j) rj[j] = reinterpret_cast<const Pack16<T>*>(xj + (int64_t)j * P)[v];
+    for (int j = 0; j < MJ; ++j) rj[j] = reinterpret_cast<const Pack16<T>*>(xj + (int64_t)j * ld)[v];
 #pragma unroll
     for (int i = 0; i < MI; ++i)
 #pragma unroll
@@ -209,9 +211,9 @@ __global__ void gram_cross_kernel(const T* __restrict__ xi, const T* __restrict_
   for (int64_t p = nvec * N + blockIdx.x * blockDim.x + threadIdx.x; p < P; p += stride) {
     float fi[MI], fj[MJ];
 #pragma unroll
-    for (int i = 0; i < MI; ++i) fi[i] = to_f(xi[(int64_t)i * P + p]);
+    for (int i = 0; i < MI; ++i) fi[i] = to_f(xi[(int64_t)i * ld + p]);
 #pragma unroll
-    for (int j = 0; j < MJ; ++j) fj[j] = to_f(xj[(int64_t)j * P + p]);
+    for (int j = 0; j < MJ; ++j) fj[j] = to_f(xj[(int64_t)j * ld + p]);
 #pragma unroll
     for (int i = 0; i < MI; ++i)
 #pragma unroll
@@ -867,23 +869,23 @@ Tensor weighted_sum(Tensor stacked, Tensor w, c10::optional<Tensor> out_opt) {
 }
 
 template <typename elem_t, int M>
-void launch_gram(const Tensor& x, Tensor& gram_flat, int64_t P) {
+void launch_gram(const Tensor& x, Tensor& gram_flat, int64_t P, int64_t ld) {
   int blocks = grid_for(P / 4, BLOCK, 2048);
   gram_kernel<elem_t, M><<<blocks, BLOCK, 0, cur_stream()>>>(
-      (const elem_t*)x.data_ptr(), P, gram_flat.data_ptr<float>());
+      (const elem_t*)x.data_ptr(), P, ld, gram_flat.data_ptr<float>());
 }
 
 // tile-pair dispatch for the decomposed path: sizes are 8 or the ragged
 // remainder r = m % 8; only (8,8), (8,r) and (r,r) combinations occur.
 template <typename elem_t>
 void launch_gram_cross(const elem_t* xi, const elem_t* xj, int mi, int mj,
-                       int64_t P, float* out, int ldo) {
+                       int64_t P, int64_t ld, float* out, int ldo) {
   int blocks = grid_for(P / 4, BLOCK, 2048);
   hipStream_t st = cur_stream();
 #define CROSS_CASE(MI, MJ)                                                       \
   if (mi == MI && mj == MJ) {                                                    \
     gram_cross_kernel<elem_t, MI, MJ>                                            \
-        <<<blocks, BLOCK, 0, st>>>(xi, xj, P, out, ldo);                         \
+        <<<blocks, BLOCK, 0, st>>>(xi, xj, P, ld, out, ldo);                     \
     return;                                                                      \
   }
   CROSS_CASE(8, 8)
@@ -895,37 +897,68 @@ void launch_gram_cross(const elem_t* xi, const elem_t* xj, int mi, int mj,
   TORCH_CHECK(false, "gram_cross: unsupported tile sizes ", mi, "x", mj);
 }
 
+// Full [m, m] Gram matrix of the rows of x (stride(1) must be 1; stride(0)
+// may exceed size(1) — chunked column views of a [m, P] buffer are accepted,
+// which is what lets exchange overlap Gram accumulation with in-flight
+// chunks).
+Tensor gram(Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.stride(1) == 1,
+              "gram: [m, n] cuda with unit inner stride");
+  TORCH_CHECK(x.scalar_type() == at::kFloat || x.scalar_type() == at::kBFloat16);
+  int m = (int)x.size(0);
+  int64_t P = x.size(1);
+  int64_t ld = x.stride(0);
+  if (m > 64) {
+    Tensor xf = x.to(at::kFloat);
+    return at::matmul(xf, xf.t());
+  }
+  if (m > 10) {
+    Tensor g = at::zeros({m, m}, x.options().dtype(at::kFloat));
+    float* gp = g.data_ptr<float>();
+    int ntiles = (m + 7) / 8;
+    DISPATCH_FT(x, {
+      const elem_t* base = (const elem_t*)x.data_ptr();
+      for (int a = 0; a < ntiles; ++a) {
+        int mi = std::min(8, m - a * 8);
+        for (int b = a; b < ntiles; ++b) {
+          int mj = std::min(8, m - b * 8);
+          launch_gram_cross<elem_t>(base + (int64_t)a * 8 * ld,
+                                    base + (int64_t)b * 8 * ld, mi, mj, P, ld,
+                                    gp + (int64_t)a * 8 * m + b * 8, m);
+        }
+      }
+    });
+    return g.triu() + g.triu(1).t();
+  }
+  int npair = m * (m + 1) / 2;
+  Tensor gram_flat = at::zeros({npair}, x.options().dtype(at::kFloat));
+  DISPATCH_FT(x, {
+    switch (m) {
+      case 1: launch_gram<elem_t, 1>(x, gram_flat, P, ld); break;
+      case 2: launch_gram<elem_t, 2>(x, gram_flat, P, ld); break;
+      case 3: launch_gram<elem_t, 3>(x, gram_flat, P, ld); break;
+      case 4: launch_gram<elem_t, 4>(x, gram_flat, P, ld); break;
+      case 5: launch_gram<elem_t, 5>(x, gram_flat, P, ld); break;
+      case 6: launch_gram<elem_t, 6>(x, gram_flat, P, ld); break;
+      case 7: launch_gram<elem_t, 7>(x, gram_flat, P, ld); break;
+      case 8: launch_gram<elem_t, 8>(x, gram_flat, P, ld); break;
+      case 9: launch_gram<elem_t, 9>(x, gram_flat, P, ld); break;
+      case 10: launch_gram<elem_t, 10>(x, gram_flat, P, ld); break;
+    }
+  });
+  Tensor g = at::zeros({m, m}, gram_flat.options());
+  auto idx = at::triu_indices(m, m, 0, gram_flat.options().dtype(at::kLong));
+  g.index_put_({idx[0], idx[1]}, gram_flat);
+  return g.triu() + g.triu(1).t();
+}
+
 Tensor pairwise_sq_dists(Tensor stacked) {
   check_flat(stacked, "stacked");
   TORCH_CHECK(stacked.dim() == 2);
   int m = (int)stacked.size(0);
   int64_t P = stacked.size(1);
-  if (m > 64) {
-    // plain library GEMM path for very large m
-    Tensor x = stacked.to(at::kFloat);
-    Tensor g = at::matmul(x, x.t());
-    Tensor sq = g.diagonal();
-    return (sq.unsqueeze(0) + sq.unsqueeze(1) - 2.0 * g).clamp_min_(0.0);
-  }
   if (m > 10) {
-    // 8x8 row-tile decomposition (monolithic kernel spills registers here)
-    Tensor gram = at::zeros({m, m}, stacked.options().dtype(at::kFloat));
-    float* gp = gram.data_ptr<float>();
-    int ntiles = (m + 7) / 8;
-    DISPATCH_FT(stacked, {
-      const elem_t* base = (const elem_t*)stacked.data_ptr();
-      for (int a = 0; a < ntiles; ++a) {
-        int mi = std::min(8, m - a * 8);
-        for (int b = a; b < ntiles; ++b) {
-          int mj = std::min(8, m - b * 8);
-          launch_gram_cross<elem_t>(base + (int64_t)a * 8 * P,
-                                    base + (int64_t)b * 8 * P, mi, mj, P,
-                                    gp + (int64_t)a * 8 * m + b * 8, m);
-        }
-      }
-    });
-    // mirror the strict upper tiles into the lower triangle
-    Tensor full = gram.triu() + gram.triu(1).t();
+    Tensor full = gram(stacked);
     Tensor sq = full.diagonal();
     return (sq.unsqueeze(0) + sq.unsqueeze(1) - 2.0 * full).clamp_min_(0.0);
   }
@@ -933,22 +966,16 @@ Tensor pairwise_sq_dists(Tensor stacked) {
   Tensor gram_flat = at::zeros({npair}, stacked.options().dtype(at::kFloat));
   DISPATCH_FT(stacked, {
     switch (m) {
-      case 1: launch_gram<elem_t, 1>(stacked, gram_flat, P); break;
-      case 2: launch_gram<elem_t, 2>(stacked, gram_flat, P); break;
-      case 3: launch_gram<elem_t, 3>(stacked, gram_flat, P); break;
-      case 4: launch_gram<elem_t, 4>(stacked, gram_flat, P); break;
-      case 5: launch_gram<elem_t, 5>(stacked, gram_flat, P); break;
-      case 6: launch_gram<elem_t, 6>(stacked, gram_flat, P); break;
-      case 7: launch_gram<elem_t, 7>(stacked, gram_flat, P); break;
-      case 8: launch_gram<elem_t, 8>(stacked, gram_flat, P); break;
-      case 9: launch_gram<elem_t, 9>(stacked, gram_flat, P); break;
-      case 10: launch_gram<elem_t, 10>(stacked, gram_flat, P); break;
-      case 11: launch_gram<elem_t, 11>(stacked, gram_flat, P); break;
-      case 12: launch_gram<elem_t, 12>(stacked, gram_flat, P); break;
-      case 13: launch_gram<elem_t, 13>(stacked, gram_flat, P); break;
-      case 14: launch_gram<elem_t, 14>(stacked, gram_flat, P); break;
-      case 15: launch_gram<elem_t, 15>(stacked, gram_flat, P); break;
-      case 16: launch_gram<elem_t, 16>(stacked, gram_flat, P); break;
+      case 1: launch_gram<elem_t, 1>(stacked, gram_flat, P, P); break;
+      case 2: launch_gram<elem_t, 2>(stacked, gram_flat, P, P); break;
+      case 3: launch_gram<elem_t, 3>(stacked, gram_flat, P, P); break;
+      case 4: launch_gram<elem_t, 4>(stacked, gram_flat, P, P); break;
+      case 5: launch_gram<elem_t, 5>(stacked, gram_flat, P, P); break;
+      case 6: launch_gram<elem_t, 6>(stacked, gram_flat, P, P); break;
+      case 7: launch_gram<elem_t, 7>(stacked, gram_flat, P, P); break;
+      case 8: launch_gram<elem_t, 8>(stacked, gram_flat, P, P); break;
+      case 9: launch_gram<elem_t, 9>(stacked, gram_flat, P, P); break;
+      case 10: launch_gram<elem_t, 10>(stacked, gram_flat, P, P); break;
     }
   });
   // unpack upper-triangular flat gram -> full [m, m] sq-dist matrix (tiny)
@@ -1273,6 +1300,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("weighted_sum", &weighted_sum, "K1: out = sum_i w_i * x_i",
         py::arg("stacked"), py::arg("w"), py::arg("out") = py::none());
   m.def("pairwise_sq_dists", &pairwise_sq_dists, "K2: [m,m] squared L2 matrix");
+  m.def("gram", &gram, "K2: [m,m] Gram matrix (accepts chunked column views)");
   m.def("row_norms", &row_norms, "K12: per-row L2 norms");
   m.def("l2_dists_to", &l2_dists_to, "K2 variant: dists of rows to own");
   m.def("count_sketch", &count_sketch, "K4: count-sketch projection");

@@ -104,3 +104,70 @@ def gather_metrics(metrics: Optional[dict], dst: int = 0) -> Optional[List[dict]
 
 def make_topology_neighbors(topo: Topology, rank: int) -> List[int]:
     return list(topo.neighbors[rank])
+
+
+def exchange_chunked_with_gram(
+    own: Tensor,
+    neighbor_ids: Sequence[int],
+    num_chunks: int = 8,
+) -> "tuple[Tensor, Tensor]":
+    """Chunked symmetric exchange overlapped with Gram accumulation.
+
+    The P-vector is split into ``num_chunks`` column chunks; all chunks'
+    sends/recvs are posted up front (RCCL progresses them on its comm
+    stream), and as each chunk's receives complete, the Gram contribution of
+    that chunk — over [own + received neighbors] — is launched on the compute
+    stream. By the time the last chunk lands, the m x m Gram (the Krum /
+    BALANCE distance input) is already computed, so the
+    distance math costs no wall time beyond the wire transfer.
+
+    Returns (stacked [1+k, P] with row 0 = own, gram [m, m] fp32).
+    """
+    from murmura_amd import ops
+
+    k = len(neighbor_ids)
+    P = own.numel()
+    m = k + 1
+    stacked = own.new_empty((m, P))
+    stacked[0].copy_(own)
+    if k == 0:
+        return stacked, ops.gram(stacked)
+
+    stage_host = _gloo_with_cuda(own)
+    bounds = [(i * P) // num_chunks for i in range(num_chunks + 1)]
+    peers = sorted(neighbor_ids)
+    row_of = {j: 1 + neighbor_ids.index(j) for j in peers}
+
+    # post ALL chunks' P2P up front; RCCL serializes them on its stream
+    reqs_per_chunk = []
+    host_bufs = []  # gloo+cuda staging
+    for c in range(num_chunks):
+        lo, hi = bounds[c], bounds[c + 1]
+        send = own[lo:hi]
+        if stage_host:
+            send = send.cpu()
+        ops_list: List[dist.P2POp] = []
+        chunk_bufs = {}
+        for j in peers:
+            recv_buf = (
+                torch.empty(hi - lo, dtype=own.dtype)
+                if stage_host
+                else stacked[row_of[j]][lo:hi]
+            )
+            chunk_bufs[j] = recv_buf
+            ops_list.append(dist.P2POp(dist.isend, send, j))
+            ops_list.append(dist.P2POp(dist.irecv, recv_buf, j))
+        reqs_per_chunk.append(dist.batch_isend_irecv(ops_list))
+        host_bufs.append(chunk_bufs)
+
+    gram_acc = torch.zeros((m, m), device=own.device, dtype=torch.float32)
+    for c in range(num_chunks):
+        for r in reqs_per_chunk[c]:
+            r.wait()
+        lo, hi = bounds[c], bounds[c + 1]
+        if stage_host:
+            for j in peers:
+                stacked[row_of[j]][lo:hi].copy_(host_bufs[c][j])
+        # chunk view [m, hi-lo] with row stride P: the Gram kernel accepts it
+        gram_acc += ops.gram(stacked[:, lo:hi])
+    return stacked, gram_acc

@@ -224,6 +224,28 @@ class FLRoundLoop:
             self._run_round_sketch_wire(own, nbr_ids, round_num)
             return
 
+        use_chunked = (
+            cfg.distributed.overlap_exchange
+            and cfg.aggregation.algorithm in ("krum", "balance")
+            and len(nbr_ids) > 0
+        )
+        if use_chunked:
+            # 3+4 overlapped: Gram accumulation (the Krum/BALANCE distance
+            # input) runs chunk-by-chunk while later chunks are in flight
+            with self.timer.phase("exchange"):
+                stacked_all, g = exchange.exchange_chunked_with_gram(own, nbr_ids)
+            with self.timer.phase("aggregate"):
+                from murmura_amd import ops as _ops
+
+                d2 = _ops.sq_dists_from_gram(g)
+                own_norm = g[0, 0].clamp_min(0).sqrt()
+                new_state = self.node.aggregate_with_neighbors(
+                    own, stacked_all[1:], neighbor_ids=nbr_ids,
+                    round_num=round_num, pairwise_d2=d2, own_norm=own_norm,
+                )
+                self.node.set_state(new_state)
+            return
+
         # 3. grouped P2P along this round's edges
         with self.timer.phase("exchange"):
             received = exchange.exchange_with_neighbors(own, nbr_ids)

@@ -192,3 +192,18 @@ def _ckpt_worker(rank, cfg_json, world, port, q, ckpt_dir, extra):
     h = run_node_process(cfg, rank, world, checkpoint_dir=ckpt_dir, **extra)
     if rank == 0:
         q.put(h)
+
+
+@pytest.mark.parametrize("algo", ["krum", "balance"])
+def test_chunked_overlap_exchange_matches_plain(algo):
+    """Chunked exchange + incremental Gram must produce the same history as
+    the plain exchange path (overlap_exchange toggles it)."""
+    base = _base_config(3, algo=algo, topo="ring", rounds=2, attack=True)
+    base["distributed"]["overlap_exchange"] = False
+    h_plain = _run_distributed(base, 3, 29650)
+    over = json.loads(json.dumps(base))
+    over["distributed"]["overlap_exchange"] = True
+    h_over = _run_distributed(over, 3, 29651)
+    for key in ["mean_accuracy", "mean_loss", "honest_accuracy"]:
+        for a, b in zip(h_over[key], h_plain[key]):
+            assert a == pytest.approx(b, abs=2e-3), (key, h_over[key], h_plain[key])
