@@ -36,7 +36,10 @@ class DistributedRunner:
                 "one node maps to one GPU in the rccl backend"
             )
         os.environ["MASTER_ADDR"] = self.config.distributed.master_addr
-        os.environ["MASTER_PORT"] = str(self.config.distributed.master_port)
+        # per-launch port offset: two concurrent runners on one host must not
+        # share a rendezvous port (cross-talk partially formed both worlds)
+        port = self.config.distributed.master_port + (os.getpid() % 997)
+        os.environ["MASTER_PORT"] = str(port)
         # avoid CPU thread oversubscription: N worker processes on one host
         if not torch.cuda.is_available():
             os.environ.setdefault("OMP_NUM_THREADS", "2")
