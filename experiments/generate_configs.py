@@ -115,9 +115,10 @@ def main() -> None:
                           topology=topo, rounds=args.rounds)
         write("topologies", f"{algo}_{topo}", cfg)
 
-    # dmtt 3-condition study
+    # dmtt 3-condition study (balance aggregation: a defended baseline so the
+    # trust protocol's collaborator selection has a measurable effect)
     for cond in ["static", "mobility", "dmtt"]:
-        cfg = base_config(f"dmtt_{cond}", "uci_har", "fedavg", args.synthetic,
+        cfg = base_config(f"dmtt_{cond}", "uci_har", "balance", args.synthetic,
                           rounds=min(args.rounds, 30))
         cfg["topology"]["type"] = "ring"
         cfg["backend"] = "distributed"
