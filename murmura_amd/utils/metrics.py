@@ -66,3 +66,10 @@ def evaluate_evidential(
         "strength": strength / n,
         "num_samples": n,
     }
+
+
+@torch.no_grad()
+def compute_accuracy(model: nn.Module, loader, device: torch.device,
+                     dtype: torch.dtype = torch.float32) -> float:
+    """Plain accuracy over a loader (reference: utils/metrics.py:50-66)."""
+    return float(evaluate_model(model, loader, device, dtype)["accuracy"])
