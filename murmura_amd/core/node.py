@@ -14,6 +14,7 @@ Differences from the reference by design:
 
 from __future__ import annotations
 
+import os
 from typing import Any, Callable, Dict, List, Optional
 
 import torch
@@ -66,8 +67,6 @@ class Node:
 
     # ------------------------------------------------------------ hipGraphs
     def _graphs_enabled(self) -> bool:
-        import os
-
         return (
             self.device.type == "cuda"
             and os.environ.get("MURMURA_NO_GRAPHS") != "1"
