@@ -60,8 +60,20 @@ def main() -> None:
     if args.limit:
         files = files[: args.limit]
     results = {}
+    # resume: keep prior successful results, re-run errors/missing
+    out_path = Path(args.out)
+    if out_path.exists():
+        try:
+            prior = json.loads(out_path.read_text())
+            results = {k: v for k, v in prior.items() if v.get("status") == "ok"}
+            if results:
+                print(f"resuming: {len(results)} prior results kept")
+        except Exception:
+            pass
     for i, f in enumerate(files):
         key = str(f.relative_to(args.configs)).replace("/", "__").removesuffix(".yaml")
+        if key in results:
+            continue
         print(f"[{i + 1}/{len(files)}] {key}", flush=True)
         try:
             results[key] = run_one(f, args.timeout)
