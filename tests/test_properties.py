@@ -81,3 +81,18 @@ def test_accept_weights_is_distribution(k, seed, thresh):
     else:
         # fallback: all weight on the closest
         assert w[dists.argmin()].item() == 1.0
+
+
+@settings(max_examples=20, deadline=None)
+@given(n=st.integers(2, 12), seed=st.integers(0, 500), r=st.integers(0, 10))
+def test_mobility_positions_and_determinism(n, seed, r):
+    from murmura_amd.topology.dynamic import MobilityModel
+
+    a = MobilityModel(n, area_size=50.0, comm_range=20.0, seed=seed)
+    b = MobilityModel(n, area_size=50.0, comm_range=20.0, seed=seed)
+    pa, pb = a.positions_at(r), b.positions_at(r)
+    assert (pa == pb).all()
+    assert (pa >= 0).all() and (pa < 50.0).all()
+    ta, tb = a.topology_at(r), b.topology_at(r)
+    assert ta.edges == tb.edges
+    assert all(ta.degree(i) >= 1 for i in range(n))  # ensure_connected default

@@ -51,3 +51,30 @@ def test_checkpoint_atomic_write(tmp_path):
     assert blob["round"] == 3
     assert 0 in blob["nodes"]
     assert not (tmp_path / "a.ckpt.tmp").exists()  # atomic rename cleaned up
+
+
+def test_device_shard_cpu_roundtrip():
+    """DeviceShard works on CPU too (shuffle gather + bounds)."""
+    from torch.utils.data import DataLoader, TensorDataset
+
+    from murmura_amd.core.gpu_round import DeviceShard
+
+    x = torch.arange(40, dtype=torch.float32).view(20, 2)
+    y = torch.arange(20)
+    shard = DeviceShard.from_loader(
+        DataLoader(TensorDataset(x, y), batch_size=8), torch.device("cpu"),
+        torch.float32,
+    )
+    assert shard.n == 20
+    out_x = torch.empty(16, 2)
+    out_y = torch.empty(16, dtype=torch.long)
+    g = torch.Generator().manual_seed(0)
+    shard.shuffled(out_x, out_y, g)
+    # every drawn row is a (x, y)-consistent pair from the shard
+    for i in range(16):
+        assert out_x[i, 0].item() == 2 * out_y[i].item()
+    # same generator state -> same permutation
+    g2 = torch.Generator().manual_seed(0)
+    out_x2 = torch.empty_like(out_x); out_y2 = torch.empty_like(out_y)
+    shard.shuffled(out_x2, out_y2, g2)
+    assert torch.equal(out_y, out_y2)
