@@ -36,17 +36,21 @@ class EvidentialTrustAggregator(Aggregator):
 
     def __init__(
         self,
-        w_a: float = 0.7,
+        w_a: float = 0.5,
         tau_u: float = 0.5,
-        penalty_factor: float = 5.0,
-        gamma_ema: float = 0.5,
-        tau_base: float = 0.5,
-        gamma: float = 0.9,
-        kappa: float = 2.0,
+        penalty_factor: float = 1.0,
+        gamma_ema: float = 0.7,
+        tau_base: float = 0.3,
+        gamma: float = 0.5,
+        kappa: float = 1.0,
         alpha_self: float = 0.5,
         max_eval_samples: int = 100,
         total_rounds: int = 50,
     ):
+        # defaults match the reference ctor (evidential_trust.py:43-58:
+        # accuracy_weight=0.5, vacuity_threshold=0.5, trust_momentum=0.7,
+        # trust_threshold=0.3, gamma=0.5, kappa=1.0, self_weight=0.5;
+        # penalty exp(-(v - tau)) i.e. factor 1.0, :295-300)
         self.w_a = float(w_a)
         self.tau_u = float(tau_u)
         self.penalty_factor = float(penalty_factor)
@@ -71,7 +75,8 @@ class EvidentialTrustAggregator(Aggregator):
         penalty = torch.exp(
             -self.penalty_factor * (vacuity - self.tau_u).clamp_min(0.0)
         )
-        return base * penalty
+        # trust clamped to [0, 1] (reference: evidential_trust.py:305)
+        return (base * penalty).clamp(0.0, 1.0)
 
     def aggregate(
         self,

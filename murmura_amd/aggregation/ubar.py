@@ -14,7 +14,6 @@ neighbor); the keep mask, fallback and blend are branchless device tensors.
 
 from __future__ import annotations
 
-import math
 from typing import Any, Dict, List
 
 import torch
@@ -63,9 +62,10 @@ class UBARAggregator(Aggregator):
             # context is provided (ubar.py:56-99 requires kwargs)
             raise ValueError("UBAR requires an eval_context (train batch + model template)")
 
-        # ---- stage 1: distance filter (device-side)
+        # ---- stage 1: distance filter (device-side); floor on rho*k
+        # matches the reference (ubar.py:135 int(self.rho * num_neighbors))
         dists = ops.l2_dists_to(own_state, neighbor_states)
-        num_keep = max(self.min_neighbors, int(math.ceil(self.rho * k)))
+        num_keep = max(self.min_neighbors, int(self.rho * k))
         num_keep = min(num_keep, k)
         _, keep_idx = torch.topk(dists, num_keep, largest=False)
         candidates = neighbor_states.index_select(0, keep_idx)

@@ -9,7 +9,6 @@ set with no communication.
 from __future__ import annotations
 
 import abc
-import math
 import random
 from typing import List
 
@@ -19,11 +18,11 @@ from torch import Tensor
 def select_compromised(
     num_nodes: int, percentage: float, seed: int
 ) -> List[int]:
-    """Seeded sample of ceil(pct * n) nodes, at least 1 when pct > 0
-    (reference: attacks/gaussian.py:37-44)."""
+    """Seeded sample of int(pct * n) nodes (floor — reference:
+    attacks/gaussian.py:37-39), at least 1 when pct > 0."""
     if percentage <= 0.0:
         return []
-    num = min(num_nodes, max(1, math.ceil(percentage * num_nodes)))
+    num = min(num_nodes, max(1, int(percentage * num_nodes)))
     rng = random.Random(seed)
     return sorted(rng.sample(range(num_nodes), num))
 

@@ -79,10 +79,16 @@ class AsyncEvaluator:
         stream; returns a handle to resolve later."""
         ready = torch.cuda.Event()
         ready.record(torch.cuda.current_stream(self.node.device))
+        copied = torch.cuda.Event()
         done = torch.cuda.Event()
         with torch.cuda.stream(self.stream):
             self.stream.wait_event(ready)
             self.store.flat.copy_(self.node.store.flat, non_blocking=True)
+            copied.record(self.stream)
             self.graph.graph.replay()
             done.record(self.stream)
+        # reverse sync: the default stream must not mutate node.store.flat
+        # until the snapshot copy has finished, or the evaluated state could
+        # be a half-mutated mix (the eval replay itself may still overlap)
+        torch.cuda.current_stream(self.node.device).wait_event(copied)
         return EvalHandle(self.graph, done, round_num)

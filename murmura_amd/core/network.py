@@ -113,8 +113,11 @@ class Network:
         """Snapshot-once, aggregate-all, apply-after-barrier
         (reference: network.py:105-139)."""
         # one pre-round snapshot per node; on a single shared device this is
-        # one stacked [N, P] tensor
-        states: List[Tensor] = [node.get_state() for node in self.nodes]
+        # one stacked [N, P] tensor. The attack alters only the BROADCAST
+        # copy — a compromised node's own aggregation uses its clean state
+        # (reference: node.py:234 own_state = self.get_state()).
+        clean: List[Tensor] = [node.get_state() for node in self.nodes]
+        states: List[Tensor] = list(clean)
         if self.attack is not None:
             for i in self.attack.get_compromised_nodes():
                 states[i] = self.attack.apply_attack(i, states[i], round_num)
@@ -129,7 +132,7 @@ class Network:
             else:
                 stacked = states[i].new_zeros((0, states[i].numel()))
             new_states[i] = node.aggregate_with_neighbors(
-                states[i], stacked, neighbor_ids=nbr_ids, round_num=round_num
+                clean[i], stacked, neighbor_ids=nbr_ids, round_num=round_num
             )
         # barrier: apply all aggregated states only after every node aggregated
         for node, ns in zip(self.nodes, new_states):

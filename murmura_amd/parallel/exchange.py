@@ -110,6 +110,7 @@ def exchange_chunked_with_gram(
     own: Tensor,
     neighbor_ids: Sequence[int],
     num_chunks: int = 8,
+    wire: Optional[Tensor] = None,
 ) -> "tuple[Tensor, Tensor]":
     """Chunked symmetric exchange overlapped with Gram accumulation.
 
@@ -122,9 +123,15 @@ def exchange_chunked_with_gram(
     distance math costs no wall time beyond the wire transfer.
 
     Returns (stacked [1+k, P] with row 0 = own, gram [m, m] fp32).
+
+    ``wire``: optional separate tensor to SEND (a compromised rank's attacked
+    broadcast copy) while ``own`` (the clean snapshot) stays row 0 of the
+    local Gram/aggregation input.
     """
     from murmura_amd import ops
 
+    if wire is None:
+        wire = own
     k = len(neighbor_ids)
     P = own.numel()
     m = k + 1
@@ -143,7 +150,7 @@ def exchange_chunked_with_gram(
     host_bufs = []  # gloo+cuda staging
     for c in range(num_chunks):
         lo, hi = bounds[c], bounds[c + 1]
-        send = own[lo:hi]
+        send = wire[lo:hi]
         if stage_host:
             send = send.cpu()
         ops_list: List[dist.P2POp] = []

@@ -12,8 +12,9 @@ Replaces the reference's ZMQ NodeProcess (murmura/distributed/node_process.py:60
 Round structure (mirrors Network.train semantics exactly so histories match
 the simulation backend on the same seeds):
   1. local train (honest nodes)
-  2. snapshot flat state; compromised ranks attack their own snapshot
-     (they broadcast AND aggregate the attacked state, reference:
+  2. snapshot flat state; compromised ranks attack only the BROADCAST copy
+     (their own aggregation uses the clean snapshot, reference: node.py:234
+     own_state = self.get_state(); the attack alters what neighbors receive,
      network.py:110-119)
   3. exchange: all-reduce fast path (fedavg + fully-connected) or grouped
      RCCL P2P along this round's edges
@@ -199,11 +200,13 @@ class FLRoundLoop:
             if not ready[self.rank]:
                 # overran the budget: skip exchange, keep own state
                 return
-        # 2. snapshot + self-attack
+        # 2. snapshot; the attack alters only the broadcast copy (wire) —
+        # own aggregation keeps the clean snapshot (reference: node.py:234)
         with self.timer.phase("snapshot_attack"):
             own = self.node.get_state()
+            wire = own
             if self._is_compromised(self.rank):
-                own = self.attack.apply_attack(self.rank, own, round_num)
+                wire = self.attack.apply_attack(self.rank, own, round_num)
 
         nbr_ids = list(topo.neighbors[self.rank])
         use_allreduce = (
@@ -216,12 +219,12 @@ class FLRoundLoop:
         if use_allreduce:
             # K1 folded into the collective: new state = global mean
             with self.timer.phase("exchange"):
-                new_state = exchange.allreduce_mean(own)
+                new_state = exchange.allreduce_mean(wire)
             self.node.set_state(new_state)
             return
 
         if self.sketch_mode and cfg.aggregation.algorithm == "sketchguard":
-            self._run_round_sketch_wire(own, nbr_ids, round_num)
+            self._run_round_sketch_wire(own, wire, nbr_ids, round_num)
             return
 
         use_chunked = (
@@ -233,7 +236,9 @@ class FLRoundLoop:
             # 3+4 overlapped: Gram accumulation (the Krum/BALANCE distance
             # input) runs chunk-by-chunk while later chunks are in flight
             with self.timer.phase("exchange"):
-                stacked_all, g = exchange.exchange_chunked_with_gram(own, nbr_ids)
+                stacked_all, g = exchange.exchange_chunked_with_gram(
+                    own, nbr_ids, wire=wire
+                )
             with self.timer.phase("aggregate"):
                 from murmura_amd import ops as _ops
 
@@ -246,9 +251,9 @@ class FLRoundLoop:
                 self.node.set_state(new_state)
             return
 
-        # 3. grouped P2P along this round's edges
+        # 3. grouped P2P along this round's edges (send the wire copy)
         with self.timer.phase("exchange"):
-            received = exchange.exchange_with_neighbors(own, nbr_ids)
+            received = exchange.exchange_with_neighbors(wire, nbr_ids)
         if nbr_ids:
             stacked = torch.stack([received[j] for j in nbr_ids], dim=0)
         else:
@@ -260,7 +265,7 @@ class FLRoundLoop:
             )
             self.node.set_state(new_state)
 
-    def _run_round_sketch_wire(self, own, nbr_ids, round_num: int) -> None:
+    def _run_round_sketch_wire(self, own, wire, nbr_ids, round_num: int) -> None:
         """Sketchguard sketch-first exchange (the comm-saving mode the
         reference left latent, sketchguard.py:114-132):
 
@@ -275,8 +280,9 @@ class FLRoundLoop:
         import torch as _t
 
         agg = self.node.aggregator
-        own_sketch = agg.get_sketch(own)  # [S]
-        wire_sk = own_sketch.contiguous()
+        own_sketch = agg.get_sketch(own)  # [S] — clean, for local filtering
+        # broadcast the sketch of the WIRE copy (what neighbors would see)
+        wire_sk = (own_sketch if wire is own else agg.get_sketch(wire)).contiguous()
         if wire_sk.is_cuda and dist.get_backend() == "gloo":
             wire_sk = wire_sk.cpu()
         all_sk = [_t.empty_like(wire_sk) for _ in range(self.world)]
@@ -305,7 +311,7 @@ class FLRoundLoop:
             ),
         }
         sym = exchange.symmetrize_wants(want, self.world, own.device)
-        received = exchange.exchange_with_neighbors(own, sym[self.rank])
+        received = exchange.exchange_with_neighbors(wire, sym[self.rank])
         use = [j for j in want if j in received]
         if use:
             stacked = _t.stack([received[j] for j in use], dim=0)

@@ -11,7 +11,8 @@ from murmura_amd.attacks import (
 def test_select_compromised_count_and_determinism():
     assert select_compromised(10, 0.0, 1) == []
     assert len(select_compromised(10, 0.2, 1)) == 2
-    assert len(select_compromised(10, 0.25, 1)) == 3  # ceil
+    assert len(select_compromised(10, 0.25, 1)) == 2  # floor (reference: gaussian.py:37)
+    assert len(select_compromised(10, 0.05, 1)) == 1  # at least 1 when pct > 0
     assert len(select_compromised(10, 0.01, 1)) == 1  # at least 1
     assert select_compromised(10, 0.3, 7) == select_compromised(10, 0.3, 7)
     assert select_compromised(10, 1.0, 1) == list(range(10))

@@ -99,7 +99,8 @@ def test_krum_resists_gaussian_attack():
 
 def test_fedavg_collapses_under_strong_attack():
     set_seed(0)
-    atk = GaussianAttack(6, 0.2, noise_std=50.0, seed=42)
+    # 34% -> floor gives 2 of 6 compromised (reference floor semantics)
+    atk = GaussianAttack(6, 0.34, noise_std=50.0, seed=42)
     net = _make_network(n=6, topo_type="fully", attack=atk)
     h = net.train(rounds=5, local_epochs=1, lr=0.1)
     # no defense: accuracy stays near chance (1/3)
@@ -177,3 +178,37 @@ def test_checkpoint_preserves_nonfloat_buffers(tmp_path):
     node.store.copy_from_flat(torch.zeros_like(node.store.flat))
     ckpt.restore_node_state(node, payload)
     assert torch.allclose(node.store.flat.float().cpu(), payload["flat"], atol=1e-6)
+
+
+def test_compromised_node_aggregates_with_clean_own_state():
+    """The attack alters only the broadcast copy; a compromised node's own
+    aggregation uses its clean snapshot (reference: node.py:234)."""
+    set_seed(0)
+    atk = GaussianAttack(4, 0.25, noise_std=1000.0, seed=42)
+    net = _make_network(n=4, topo_type="ring", attack=atk)
+    comp = atk.get_compromised_nodes()[0]
+    honest = [i for i in range(4) if i != comp]
+
+    clean_states = [n.get_state().clone() for n in net.nodes]
+    net._aggregation_step(0, net.topology)
+
+    # FedAvg over a ring: node i's new state = mean of {own, two neighbors}.
+    # For the compromised node, "own" must be the CLEAN state even though the
+    # neighbors of the compromised node received an attacked copy.
+    nbrs = net.topology.neighbors[comp]
+    expected = torch.stack(
+        [clean_states[comp]]
+        + [atk.apply_attack(j, clean_states[j], 0) if atk.is_compromised(j)
+           else clean_states[j] for j in nbrs]
+    ).mean(dim=0)
+    got = net.nodes[comp].get_state()
+    # σ=1000 noise would dominate; closeness proves the clean state was used
+    assert torch.allclose(got, expected, atol=1e-4)
+
+    # honest neighbors of the compromised node DID receive the attacked copy
+    h = [i for i in honest if comp in net.topology.neighbors[i]][0]
+    nbrs_h = net.topology.neighbors[h]
+    with_clean = torch.stack(
+        [clean_states[h]] + [clean_states[j] for j in nbrs_h]
+    ).mean(dim=0)
+    assert not torch.allclose(net.nodes[h].get_state(), with_clean, atol=1.0)
