@@ -101,6 +101,13 @@ class DistributedConfig(_Strict):
     The reference used ZMQ endpoints + wall-clock rounds
     (reference: config/schema.py:7-51); the RCCL backend replaces the transport
     but keeps the knobs that still make sense, plus rendezvous settings.
+
+    Reference YAMLs carrying ZMQ-era keys (``transport``, ``ipc_dir``,
+    ``host``, ``coordinator_*_port``, ``base_port``, ``node_hosts``,
+    ``startup_grace_s``) are ACCEPTED and ignored with a warning — the YAML
+    schema is an explicit compatibility surface (extend
+    additively without breaking existing keys). ``round_duration_s`` maps
+    through to the straggler budget.
     """
 
     # rendezvous for torch.distributed (one process per node/GPU)
@@ -109,7 +116,9 @@ class DistributedConfig(_Strict):
     # communication backend: "nccl" is RCCL on ROCm; "gloo" for CPU tests
     comm_backend: Literal["auto", "nccl", "gloo"] = "auto"
     # straggler semantics: intra-box RCCL is deterministic-synchronous; a
-    # wall-clock round budget is kept for parity experiments only (0 = off)
+    # wall-clock round budget is kept for parity experiments only (0 = off).
+    # Maps 1:1 from the reference's wall-clock round window
+    # (reference: config/schema.py:46-51).
     round_duration_s: float = 0.0
     # overlap neighbor exchange with eval/scoring compute on a side stream
     overlap_exchange: bool = True
@@ -117,6 +126,36 @@ class DistributedConfig(_Strict):
     # (sketchguard wire-compression mode; reference kept it latent,
     # sketchguard.py:114-132)
     sketch_wire_mode: bool = False
+
+    # ---- ZMQ-era keys (reference: config/schema.py:10-51): accepted for
+    # compatibility, IGNORED by the RCCL backend (warned at load time).
+    transport: Optional[Literal["ipc", "tcp"]] = None
+    ipc_dir: Optional[str] = None
+    host: Optional[str] = None
+    coordinator_pub_port: Optional[int] = None
+    coordinator_pull_port: Optional[int] = None
+    base_port: Optional[int] = None
+    node_hosts: Optional[Dict[int, str]] = None
+    startup_grace_s: Optional[float] = None
+
+    _ZMQ_LEGACY_KEYS = (
+        "transport", "ipc_dir", "host", "coordinator_pub_port",
+        "coordinator_pull_port", "base_port", "node_hosts", "startup_grace_s",
+    )
+
+    @model_validator(mode="after")
+    def _warn_legacy(self) -> "DistributedConfig":
+        legacy = [k for k in self._ZMQ_LEGACY_KEYS if getattr(self, k) is not None]
+        if legacy:
+            import warnings
+
+            warnings.warn(
+                f"distributed: ZMQ-era keys {legacy} are accepted for "
+                "reference-YAML compatibility but ignored by the RCCL backend",
+                UserWarning,
+                stacklevel=2,
+            )
+        return self
 
 
 class ComputeConfig(_Strict):

@@ -14,6 +14,10 @@ from murmura_amd.models.evidential import (
 )
 
 _NAMES = {
+    # reference factory names (reference: models.py:447-454)
+    "uci_har": EvidentialHARClassifier,
+    "pamap2": EvidentialPAMAP2Classifier,
+    "ppg_dalia": EvidentialPPGDaLiAClassifier,
     "har_classifier": EvidentialHARClassifier,
     "evidential_har": EvidentialHARClassifier,
     "pamap2_classifier": EvidentialPAMAP2Classifier,
@@ -24,7 +28,7 @@ _NAMES = {
 
 
 def get_factory(name: str, **params) -> Callable[[], nn.Module]:
-    key = name.lower()
+    key = name.lower().replace("-", "_")
     if key not in _NAMES:
         raise ValueError(f"unknown wearables model {name!r}; options: {sorted(set(_NAMES))}")
     cls = _NAMES[key]

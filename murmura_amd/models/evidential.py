@@ -102,48 +102,67 @@ def _mlp_block(in_f: int, out_f: int, dropout: float = 0.3) -> nn.Sequential:
     )
 
 
-class EvidentialHARClassifier(nn.Module):
-    """UCI HAR: MLP 561-256-128-EDL6 (reference: models.py:187-243)."""
+class _EvidentialMLP(nn.Module):
+    """MLP body of ``hidden_dims`` blocks + evidential head. Accepts both the
+    reference's ctor names (``input_dim``/``hidden_dims``/``dropout``,
+    reference: models.py:355-360) and our ``in_features`` alias so verbatim
+    reference YAMLs construct models unchanged."""
 
-    def __init__(self, in_features: int = 561, num_classes: int = 6):
+    def __init__(
+        self,
+        input_dim: int,
+        hidden_dims=(256, 128),
+        num_classes: int = 6,
+        dropout: float = 0.3,
+    ):
         super().__init__()
+        dims = [input_dim] + list(hidden_dims)
         self.body = nn.Sequential(
-            _mlp_block(in_features, 256), _mlp_block(256, 128)
+            *[_mlp_block(dims[i], dims[i + 1], dropout) for i in range(len(dims) - 1)]
         )
-        self.head = EvidentialHead(128, num_classes)
+        self.head = EvidentialHead(dims[-1], num_classes)
 
     def forward(self, x: Tensor) -> Tensor:
         return self.head(self.body(x.flatten(1)))
 
 
-class EvidentialPAMAP2Classifier(nn.Module):
+def _alias_input_dim(kwargs: dict) -> dict:
+    if "in_features" in kwargs:
+        kwargs = dict(kwargs)
+        kwargs["input_dim"] = kwargs.pop("in_features")
+    return kwargs
+
+
+class EvidentialHARClassifier(_EvidentialMLP):
+    """UCI HAR: MLP 561-256-128-EDL6 (reference: models.py:187-243,355-367)."""
+
+    def __init__(self, input_dim: int = 561, hidden_dims=(256, 128),
+                 num_classes: int = 6, dropout: float = 0.3, **alias):
+        super().__init__(**_alias_input_dim(
+            dict(input_dim=input_dim, hidden_dims=hidden_dims,
+                 num_classes=num_classes, dropout=dropout, **alias)))
+
+
+class EvidentialPAMAP2Classifier(_EvidentialMLP):
     """PAMAP2: windows 100x40 -> 4000-512-256-128-EDL12
     (reference: models.py:246-286)."""
 
-    def __init__(self, in_features: int = 4000, num_classes: int = 12):
-        super().__init__()
-        self.body = nn.Sequential(
-            _mlp_block(in_features, 512), _mlp_block(512, 256), _mlp_block(256, 128)
-        )
-        self.head = EvidentialHead(128, num_classes)
-
-    def forward(self, x: Tensor) -> Tensor:
-        return self.head(self.body(x.flatten(1)))
+    def __init__(self, input_dim: int = 4000, hidden_dims=(512, 256, 128),
+                 num_classes: int = 12, dropout: float = 0.3, **alias):
+        super().__init__(**_alias_input_dim(
+            dict(input_dim=input_dim, hidden_dims=hidden_dims,
+                 num_classes=num_classes, dropout=dropout, **alias)))
 
 
-class EvidentialPPGDaLiAClassifier(nn.Module):
+class EvidentialPPGDaLiAClassifier(_EvidentialMLP):
     """PPG-DaLiA: windows 32x6 -> 192-256-128-64-EDL7
     (reference: models.py:289-347)."""
 
-    def __init__(self, in_features: int = 192, num_classes: int = 7):
-        super().__init__()
-        self.body = nn.Sequential(
-            _mlp_block(in_features, 256), _mlp_block(256, 128), _mlp_block(128, 64)
-        )
-        self.head = EvidentialHead(64, num_classes)
-
-    def forward(self, x: Tensor) -> Tensor:
-        return self.head(self.body(x.flatten(1)))
+    def __init__(self, input_dim: int = 192, hidden_dims=(256, 128, 64),
+                 num_classes: int = 7, dropout: float = 0.3, **alias):
+        super().__init__(**_alias_input_dim(
+            dict(input_dim=input_dim, hidden_dims=hidden_dims,
+                 num_classes=num_classes, dropout=dropout, **alias)))
 
 
 def get_evidential_loss(

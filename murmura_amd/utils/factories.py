@@ -108,12 +108,43 @@ def build_aggregator_factory(config: Config, model_factory=None) -> Callable[[in
     # accept the reference configs' "f" alias for krum's num_compromised
     if algo == "krum" and "f" in params:
         params["num_compromised"] = params.pop("f")
+    # reference-native parameter names for evidential_trust
+    # (reference: evidential_trust.py:43-58) -> our names
+    if algo == "evidential_trust":
+        for ref_name, ours in (
+            ("vacuity_threshold", "tau_u"),
+            ("accuracy_weight", "w_a"),
+            ("trust_threshold", "tau_base"),
+            ("self_weight", "alpha_self"),
+            ("trust_momentum", "gamma_ema"),
+        ):
+            if ref_name in params:
+                params[ours] = params.pop(ref_name)
     if algo in ("balance", "sketchguard", "ubar", "evidential_trust"):
         params.setdefault("total_rounds", config.experiment.rounds)
     if algo == "sketchguard" and "model_dim" not in params:
         if model_factory is None:
             model_factory = build_model_factory(config)
         params["model_dim"] = calculate_model_dimension(model_factory())
+    # The reference aggregator ctors all take **kwargs and silently drop
+    # unknown params (e.g. krum.py:15 ignores the configs' "m",
+    # balance ignores "threshold_multiplier", sketchguard ignores
+    # "num_buckets"/"num_hashes"). Match that YAML-level leniency here —
+    # with a warning — while keeping our ctors strict for programmatic use.
+    import inspect
+    import warnings
+
+    accepted = set(inspect.signature(cls.__init__).parameters) - {"self"}
+    unknown = [k for k in params if k not in accepted]
+    if unknown:
+        warnings.warn(
+            f"aggregation.params: ignoring unknown keys {unknown} for "
+            f"{algo!r} (the reference ctor ignores them too via **kwargs)",
+            UserWarning,
+            stacklevel=2,
+        )
+        for k in unknown:
+            params.pop(k)
     return lambda node_id: cls(**params)
 
 
