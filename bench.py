@@ -38,24 +38,57 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--preset", type=int, default=0, choices=[0, 1, 2, 3, 4, 5],
+                   help="BASELINE.json config presets: 1=2-node ring FedAvg tiny "
+                        "MLP (CPU plumbing), 2=fully FedAvg resnet18 bf16 "
+                        "(default), 3=k-regular(4) Krum + 20%% Gaussian, "
+                        "4=Sketchguard ~100M sketch-wire, 5=mobility G^t + "
+                        "DMTT/UBAR under topology-liar")
     p.add_argument("--algo", default="fedavg",
                    choices=["fedavg", "krum", "balance", "sketchguard", "ubar",
                             "evidential_trust"])
     p.add_argument("--topology", default=None,
                    help="default: fully (fedavg) / k-regular (krum)")
-    p.add_argument("--attack", default="none", choices=["none", "gaussian", "directed"])
+    p.add_argument("--attack", default="none",
+                   choices=["none", "gaussian", "directed", "topology_liar"])
     p.add_argument("--model", default="resnet18",
                    choices=["resnet18", "femnist-baseline", "femnist-xlarge", "mlp",
-                            "wide100m"])
+                            "wide100m", "har"])
     p.add_argument("--sketch-wire", action="store_true",
                    help="sketchguard: exchange 4KB sketches first, full states only with accepted neighbors")
+    p.add_argument("--mobility", action="store_true",
+                   help="dynamic topology G^t (bounded random walk, seeded)")
+    p.add_argument("--dmtt", action="store_true",
+                   help="DMTT trust protocol (implies --mobility)")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--shard", type=int, default=2048, help="samples per node")
     p.add_argument("--batch-size", type=int, default=64)
     p.add_argument("--local-epochs", type=int, default=1)
     p.add_argument("--no-eval", action="store_true",
                    help="skip per-round evaluation inside the timed region")
-    return p.parse_args()
+    args = p.parse_args()
+    return apply_preset(args)
+
+
+def apply_preset(args):
+    """BASELINE.json's five named configs as one flag each."""
+    if args.preset == 1:
+        args.algo, args.topology, args.model = "fedavg", "ring", "mlp"
+        args.dtype = "fp32"  # CPU plumbing config
+    elif args.preset == 2:
+        args.algo, args.topology, args.model = "fedavg", "fully", "resnet18"
+    elif args.preset == 3:
+        args.algo, args.topology, args.model = "krum", "k-regular", "resnet18"
+        args.attack = "gaussian"
+    elif args.preset == 4:
+        args.algo, args.model = "sketchguard", "wide100m"
+        args.sketch_wire = True
+    elif args.preset == 5:
+        args.algo, args.model = "ubar", "har"
+        args.attack, args.dmtt = "topology_liar", True
+    if args.dmtt:
+        args.mobility = True
+    return args
 
 
 def build_config(args, world):
@@ -75,20 +108,33 @@ def build_config(args, world):
         "wide100m": ("models.widemlp",
                      {"in_features": 4096, "hidden": 12288, "num_classes": 62},
                      {"num_features": 4096, "num_classes": 62}),
+        # UCI-HAR-shaped evidential MLP (BASELINE.json config 5 / DMTT)
+        "har": ("examples.wearables.uci_har", {"input_dim": 561, "num_classes": 6},
+                {"num_features": 561, "num_classes": 6}),
     }
     factory, mparams, dparams = model_cfgs[args.model]
     topo = args.topology or ("k-regular" if args.algo == "krum" else "fully")
     attack_cfg = {"enabled": False}
     if args.attack != "none":
+        atk_type = {"gaussian": "gaussian", "directed": "directed_deviation",
+                    "topology_liar": "topology_liar"}[args.attack]
         attack_cfg = {
             "enabled": True,
-            "type": "gaussian" if args.attack == "gaussian" else "directed_deviation",
+            "type": atk_type,
             "percentage": 0.2,
-            "params": {"noise_std": 10.0},
+            "params": ({"model_attack_type": "gaussian", "noise_std": 10.0}
+                       if atk_type == "topology_liar" else {"noise_std": 10.0}),
         }
     agg_params = {}
     if args.algo == "krum":
         agg_params = {"num_compromised": max(1, int(0.2 * world))}
+    extra = {}
+    if args.mobility:
+        extra["mobility"] = {"area_size": 100.0, "comm_range": 60.0,
+                             "max_speed": 8.0, "seed": 42,
+                             "ensure_connected": True}
+    if args.dmtt:
+        extra["dmtt"] = {}
     return Config(**{
         "experiment": {"name": f"bench-{args.algo}", "seed": 42,
                        "rounds": args.steps + args.warmup, "verbose": False},
@@ -104,6 +150,7 @@ def build_config(args, world):
         "backend": "rccl",
         "distributed": {"sketch_wire_mode": bool(getattr(args, "sketch_wire", False))},
         "compute": {"dtype": args.dtype, "native_kernels": True},
+        **extra,
     })
 
 
@@ -126,6 +173,7 @@ def main():
 
     import torch.distributed as dist
 
+    from murmura_amd.parallel import exchange
     from murmura_amd.parallel.node_process import FLRoundLoop, init_distributed
 
     if launched_distributed or world > 1:
@@ -140,8 +188,14 @@ def main():
         device = torch.device("cuda:0" if use_cuda else "cpu")
         if use_cuda:
             torch.cuda.set_device(device)
+        exchange.create_host_group_if_needed()
 
-    loop = FLRoundLoop(config, rank, world, device)
+    if config.dmtt is not None:
+        from murmura_amd.dmtt.node_process import DMTTRoundLoop
+
+        loop = DMTTRoundLoop(config, rank, world, device)
+    else:
+        loop = FLRoundLoop(config, rank, world, device)
 
     pending = [None]
 
@@ -162,7 +216,7 @@ def main():
     def sync():
         if use_cuda:
             torch.cuda.synchronize()
-        dist.barrier()
+        exchange.barrier(device)
         if use_cuda:
             torch.cuda.synchronize()
 
@@ -221,7 +275,9 @@ def main():
             },
         }
         print(json.dumps(result))
-    dist.barrier()
+        if getattr(loop, "last_wire_stats", None):
+            print(f"sketch-wire stats: {loop.last_wire_stats}", file=sys.stderr)
+    exchange.barrier(device)
     dist.destroy_process_group()
 
 

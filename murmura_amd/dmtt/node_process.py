@@ -50,13 +50,22 @@ class DMTTRoundLoop(FLRoundLoop):
     def _score_neighbor_models(self, received: Dict[int, torch.Tensor]) -> None:
         """Load each received flat state into the scratch model and score
         accuracy + vacuity on local data (reference: dmtt/node_process.py:309-363
-        did a fresh model + deepcopy per neighbor; here it is one flat copy)."""
+        did a fresh model + deepcopy per neighbor; here it is one flat copy).
+
+        All candidates are scored with the metrics left ON DEVICE, then read
+        back in ONE host sync (the reference's per-neighbor ``float()`` pulls
+        forced m syncs per round)."""
         ctx = self.node._get_eval_context()
         if ctx is None:
             return
-        for j, flat in received.items():
-            vac, acc = ctx.evidential_score(flat, max_samples=100)
-            self.dmtt_state.record_model_score(j, float(vac), float(acc))
+        order = list(received.keys())
+        if not order:
+            return
+        pairs = [ctx.evidential_score(received[j], max_samples=100) for j in order]
+        flat = torch.stack([t for vac_acc in pairs for t in vac_acc])
+        vals = flat.tolist()  # single host sync for all neighbors
+        for idx, j in enumerate(order):
+            self.dmtt_state.record_model_score(j, vals[2 * idx], vals[2 * idx + 1])
 
     def _process_topo_claims(
         self, claims: Dict[int, List[int]], topo, round_num: int
@@ -93,20 +102,23 @@ class DMTTRoundLoop(FLRoundLoop):
             self.node.local_train(
                 epochs=cfg.training.local_epochs, lr=cfg.training.lr, round_num=round_num
             )
-        # 2. snapshot + self-attack (topology_liar may wrap a model attack)
+        # 2. snapshot; attack alters only the broadcast copy (own aggregation
+        # keeps the clean snapshot, reference: node.py:234)
         own = self.node.get_state()
+        wire = own
         if self._is_compromised(self.rank):
-            own = self.attack.apply_attack(self.rank, own, round_num)
+            wire = self.attack.apply_attack(self.rank, own, round_num)
 
         # 3a. symmetrize the asymmetric collaborator sets -> exchange plan
-        sym_sets = exchange.symmetrize_wants(collaborators, self.world, self.device)
+        sym_sets = exchange.symmetrize_wants(collaborators, self.world)
         peers = sym_sets[self.rank]
-        received = exchange.exchange_with_neighbors(own, peers)
+        received = exchange.exchange_with_neighbors(wire, peers)
 
-        # 3b. topology claims: tiny host-side all-gather
+        # 3b. topology claims: tiny host-side all-gather (host group — under
+        # nccl an object collective would otherwise run on the device)
         claim = self._claim_for(g_neighbors, round_num)
         all_claims: List[Optional[List[int]]] = [None] * self.world
-        dist.all_gather_object(all_claims, claim)
+        dist.all_gather_object(all_claims, claim, group=exchange.host_group())
         claims = {j: all_claims[j] for j in range(self.world) if all_claims[j] is not None}
 
         # 4. trust updates

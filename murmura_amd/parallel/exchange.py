@@ -35,6 +35,43 @@ def _gloo_with_cuda(t: Tensor) -> bool:
     return t.is_cuda and dist.get_backend() == "gloo"
 
 
+# ---------------------------------------------------------------- host group
+# Object collectives (gather_object/all_gather_object) under nccl pickle to
+# uint8 CUDA tensors and run DEVICE collectives — adding device syncs to the
+# metrics path and exercising nccl semantics the gloo CPU tests never see.
+# All host-side/tiny-metadata collectives are therefore routed through a gloo
+# SUBGROUP when the main backend is nccl (created collectively in
+# init_distributed; every rank participates). Under gloo the default group is
+# already host-side and _host_group stays None.
+_host_group = None
+
+
+def set_host_group(group) -> None:
+    global _host_group
+    _host_group = group
+
+
+def host_group():
+    """Group for host-side collectives: the gloo subgroup under nccl, else
+    the default group (None)."""
+    return _host_group
+
+
+def create_host_group_if_needed() -> None:
+    """Collective — every rank must call (done in init_distributed)."""
+    if dist.get_backend() != "gloo" and _host_group is None:
+        set_host_group(dist.new_group(backend="gloo"))
+
+
+def barrier(device=None) -> None:
+    """Backend-aware barrier: under nccl, bind to this rank's device
+    explicitly instead of relying on the current-context guess."""
+    if device is not None and device.type == "cuda" and dist.get_backend() == "nccl":
+        dist.barrier(device_ids=[device.index])
+    else:
+        dist.barrier()
+
+
 def exchange_with_neighbors(
     own: Tensor, neighbor_ids: Sequence[int], tag_base: int = 0
 ) -> Dict[int, Tensor]:
@@ -69,21 +106,21 @@ def allreduce_mean(own: Tensor) -> Tensor:
     return out.to(own.device) if out.device != own.device else out
 
 
-def symmetrize_wants(want: Sequence[int], world_size: int, device) -> List[List[int]]:
+def symmetrize_wants(want: Sequence[int], world_size: int, device=None) -> List[List[int]]:
     """All-gather each rank's desired peer list and return the SYMMETRIC
     exchange sets: i exchanges with j iff i wants j OR j wants i.
 
     DMTT's collaborator sets are asymmetric; RCCL P2P requires both ends to
     post matching ops, so the union set is agreed via one tiny all-gather of
-    an N-bit mask (8 bytes on an 8-GPU box)."""
-    if dist.get_backend() == "gloo":
-        device = torch.device("cpu")
-    mask = torch.zeros(world_size, dtype=torch.uint8, device=device)
+    an N-bit mask (8 bytes on an 8-GPU box). The want lists originate on the
+    host and the result is consumed on the host, so the mask goes over the
+    host group (gloo subgroup under nccl) — no H2D/D2H round trip."""
+    mask = torch.zeros(world_size, dtype=torch.uint8)
     for j in want:
         mask[j] = 1
     all_masks = [torch.zeros_like(mask) for _ in range(world_size)]
-    dist.all_gather(all_masks, mask)
-    m = torch.stack(all_masks).cpu()  # [N, N]; m[i][j] = i wants j
+    dist.all_gather(all_masks, mask, group=host_group())
+    m = torch.stack(all_masks)  # [N, N]; m[i][j] = i wants j
     sym = m | m.t()
     out: List[List[int]] = []
     for i in range(world_size):
@@ -98,7 +135,9 @@ def gather_metrics(metrics: Optional[dict], dst: int = 0) -> Optional[List[dict]
     gathered: Optional[List[Optional[dict]]] = (
         [None] * world if dist.get_rank() == dst else None
     )
-    dist.gather_object(metrics, gathered, dst=dst)
+    # object collectives stay on the host group: under nccl they would
+    # otherwise pickle into CUDA tensors and sync the device per round
+    dist.gather_object(metrics, gathered, dst=dst, group=host_group())
     return gathered  # type: ignore[return-value]
 
 

@@ -72,6 +72,9 @@ def init_distributed(config: Config, rank: int, world_size: int) -> torch.device
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")
+    # gloo subgroup for host-side object/metadata collectives (collective —
+    # every rank calls it here; see exchange.host_group)
+    exchange.create_host_group_if_needed()
     return device
 
 
@@ -171,13 +174,13 @@ class FLRoundLoop:
         budget = self.config.distributed.round_duration_s
         if budget <= 0:
             return None
+        # host-side values in, host-side decision out: one byte per rank over
+        # the host group (gloo subgroup under nccl) — no device round trip
         ready = torch.tensor(
             [1 if train_seconds <= budget else 0], dtype=torch.uint8
         )
-        if self.device.type == "cuda" and dist.get_backend() != "gloo":
-            ready = ready.to(self.device)
         out = [torch.zeros_like(ready) for _ in range(self.world)]
-        dist.all_gather(out, ready)
+        dist.all_gather(out, ready, group=exchange.host_group())
         return [bool(t.item()) for t in out]
 
     def run_round(self, round_num: int) -> None:
@@ -466,7 +469,7 @@ def run_node_process(
     all_stats = exchange.gather_metrics(stats, dst=0)
     if rank == 0 and history is not None:
         history["node_statistics"] = {i: s for i, s in enumerate(all_stats)}
-    dist.barrier()
+    exchange.barrier(device)
     if destroy_group:
         dist.destroy_process_group()
     return history
