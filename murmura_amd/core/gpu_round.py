@@ -111,27 +111,37 @@ class TrainGraph:
         self.loss_sum += loss.detach()
 
     def _capture(self) -> None:
+        from murmura_amd.ops.fused_bn import MurmuraBatchNorm2d
+
         node = self.node
         node.model.train()
         node.store.ensure_grads()
         # warmup on a side stream executes real steps — snapshot/restore the
         # flat state so capture-time warmup does not perturb training
         saved = node.store.flat.clone()
-        s = torch.cuda.Stream(device=self.shard.device)
-        s.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(s):
-            for _ in range(2):
+        # num_batches_tracked increments are deferred out of the graph (a
+        # ~5 us device add per BN per step); local_train bumps them host-side
+        # per epoch instead (ops/fused_bn.py bump_num_batches_tracked)
+        prev_defer = MurmuraBatchNorm2d.defer_num_batches_tracked
+        MurmuraBatchNorm2d.defer_num_batches_tracked = True
+        try:
+            s = torch.cuda.Stream(device=self.shard.device)
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    self.loss_sum.zero_()
+                    self._one_step(0)
+            torch.cuda.current_stream().wait_stream(s)
+            node.store.flat.copy_(saved)
+            node.store.zero_grad()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
                 self.loss_sum.zero_()
-                self._one_step(0)
-        torch.cuda.current_stream().wait_stream(s)
-        node.store.flat.copy_(saved)
-        node.store.zero_grad()
-        g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
-            self.loss_sum.zero_()
-            for step in range(self.num_batches):
-                self._one_step(step)
-        self.graph = g
+                for step in range(self.num_batches):
+                    self._one_step(step)
+            self.graph = g
+        finally:
+            MurmuraBatchNorm2d.defer_num_batches_tracked = prev_defer
 
     def run_epoch(self, lr: float, round_num: int = 0) -> Tensor:
         """Shuffle + replay one epoch; returns the summed loss (device scalar)."""

@@ -113,6 +113,11 @@ class Node:
                 for _ in range(max(1, epochs)):
                     total = total + self._train_graph.run_epoch(lr, round_num)
                 nb = self._train_graph.num_batches * max(1, epochs)
+                # replay the deferred num_batches_tracked increments once per
+                # call (the captured graph omits the per-step device adds)
+                from murmura_amd.ops.fused_bn import bump_num_batches_tracked
+
+                bump_num_batches_tracked(self.model, nb)
                 return {"loss": float(total.item() / nb), "num_batches": nb}
         self.model.train()
         gflat = self.store.ensure_grads()

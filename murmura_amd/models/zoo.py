@@ -13,7 +13,11 @@ from typing import Callable, Dict
 import torch
 from torch import Tensor, nn
 
-from murmura_amd.ops.fused_bn import MurmuraBatchNorm2d, MurmuraBNReLU
+from murmura_amd.ops.fused_bn import (
+    MurmuraBatchNorm2d,
+    MurmuraBNAddReLU,
+    MurmuraBNReLU,
+)
 
 
 class SimpleMLP(nn.Module):
@@ -109,7 +113,9 @@ class BasicBlock(nn.Module):
         self.conv1 = nn.Conv2d(in_ch, out_ch, 3, stride=stride, padding=1, bias=False)
         self.bn1 = MurmuraBNReLU(out_ch)  # ReLU folded into the BN kernels
         self.conv2 = nn.Conv2d(out_ch, out_ch, 3, stride=1, padding=1, bias=False)
-        self.bn2 = MurmuraBatchNorm2d(out_ch)
+        # block tail relu(bn2(conv2) + shortcut) fused into one kernel each
+        # direction (residual grad comes from the same backward launch)
+        self.bn2 = MurmuraBNAddReLU(out_ch)
         self.shortcut = nn.Sequential()
         if stride != 1 or in_ch != out_ch:
             self.shortcut = nn.Sequential(
@@ -119,8 +125,7 @@ class BasicBlock(nn.Module):
 
     def forward(self, x: Tensor) -> Tensor:
         out = self.bn1(self.conv1(x))  # BN+ReLU fused
-        out = self.bn2(self.conv2(out))
-        return torch.relu(out + self.shortcut(x))
+        return self.bn2(self.conv2(out), res=self.shortcut(x))
 
 
 class ResNet18(nn.Module):

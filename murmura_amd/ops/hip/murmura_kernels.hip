@@ -586,96 +586,199 @@ __global__ void evidential_stats_kernel(const T* __restrict__ logits,
 // through one LDS histogram pass, then one global atomic per channel per
 // block. (The first version assigned one scalar channel per thread — 2-byte
 // loads, 190 GB/s; packs restore the streaming roofline.)
-template <typename T>
-__global__ void bn_sums_kernel(const T* __restrict__ x, int64_t R, int C,
-                               float* __restrict__ sum, float* __restrict__ sumsq) {
+// Round-2 redesign. Round 1 capped the reduction grid at 64 blocks (global
+// atomics serialized beyond that) which left 192 of 256 CUs idle and ran at
+// ~0.6 TB/s, and every call pre-zeroed a [2,C] workspace (a ~4.8 us
+// FillFunctor launch x 40/step). New scheme:
+//   phase 1  bn_partials_kernel   per-block partials, NO atomics, NO
+//                                 pre-zeroing, grid sized for the chip
+//   finalize bn_finalize_*_kernel tiny: reduces G partials AND precomputes
+//                                 the per-channel coefficients (removes the
+//                                 per-block recompute from the elementwise
+//                                 pass)
+//   pass 2   bn_norm/bn_bwd_dx    elementwise; optionally fuses the ResNet
+//                                 residual add (+ReLU) and its backward,
+//                                 removing 2 elementwise kernels per block
+//                                 per direction
+// FWD accumulates (x, x^2); BWD accumulates (g, g*xhat) with the ReLU mask
+// from the saved output (mask applies to the residual grad too, since the
+// ReLU follows the add).
+template <typename T, bool BWD, bool RELU>
+__global__ void bn_partials_kernel(const T* __restrict__ x,
+                                   const T* __restrict__ dy,
+                                   const T* __restrict__ yout,
+                                   int64_t R, int C,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ invstd,
+                                   float* __restrict__ ws) {
   constexpr int N = Pack16<T>::N;
-  const int packs_per_row = C / N;          // caller guarantees divisibility
-  const int pk = threadIdx.x % packs_per_row;
-  const int rsub = threadIdx.x / packs_per_row;
-  const int rows_per_iter = blockDim.x / packs_per_row;
+  const int ppr = C / N;  // packs per row; caller guarantees divisibility
+  const int pk = threadIdx.x % ppr;
+  const int rsub = threadIdx.x / ppr;
+  const int rows_per_iter = max(1, (int)blockDim.x / ppr);
   const int cbase = pk * N;
 
-  float sacc[N], qacc[N];
+  float sacc[N], qacc[N], m[N], is[N];
 #pragma unroll
-  for (int k = 0; k < N; ++k) sacc[k] = qacc[k] = 0.0f;
-
+  for (int k = 0; k < N; ++k) {
+    sacc[k] = qacc[k] = 0.0f;
+    if (BWD) {
+      m[k] = mean[cbase + k];
+      is[k] = invstd[cbase + k];
+    }
+  }
   if (rsub < rows_per_iter) {
     for (int64_t r = (int64_t)blockIdx.x * rows_per_iter + rsub; r < R;
          r += (int64_t)gridDim.x * rows_per_iter) {
-      Pack16<T> v = *reinterpret_cast<const Pack16<T>*>(x + r * C + cbase);
+      Pack16<T> xv = *reinterpret_cast<const Pack16<T>*>(x + r * C + cbase);
+      if (!BWD) {
 #pragma unroll
-      for (int k = 0; k < N; ++k) {
-        float f = to_f(v.e[k]);
-        sacc[k] += f;
-        qacc[k] = fmaf(f, f, qacc[k]);
+        for (int k = 0; k < N; ++k) {
+          float f = to_f(xv.e[k]);
+          sacc[k] += f;
+          qacc[k] = fmaf(f, f, qacc[k]);
+        }
+      } else {
+        Pack16<T> gv = *reinterpret_cast<const Pack16<T>*>(dy + r * C + cbase);
+        Pack16<T> yv;
+        if (RELU) yv = *reinterpret_cast<const Pack16<T>*>(yout + r * C + cbase);
+#pragma unroll
+        for (int k = 0; k < N; ++k) {
+          float g = to_f(gv.e[k]);
+          if (RELU && to_f(yv.e[k]) <= 0.0f) g = 0.0f;
+          float xh = (to_f(xv.e[k]) - m[k]) * is[k];
+          sacc[k] += g;
+          qacc[k] = fmaf(g, xh, qacc[k]);
+        }
       }
     }
   }
-  extern __shared__ float lds[];  // csum[C], csq[C]
-  float* csum = lds;
-  float* csq = lds + C;
-  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.0f;
-  __syncthreads();
+  // block combine -> ws[blockIdx][0..C)=sum, [C..2C)=sumsq (no atomics on
+  // the pow2 path: cross-lane shfl tree, then one LDS row per wave)
+  extern __shared__ float lds[];
+  float* out = ws + (int64_t)blockIdx.x * 2 * C;
+  const int nwaves = blockDim.x / WAVE;
+  if ((ppr & (ppr - 1)) == 0 && ppr <= WAVE) {
 #pragma unroll
-  for (int k = 0; k < N; ++k) {
-    atomicAdd(&csum[cbase + k], sacc[k]);
-    atomicAdd(&csq[cbase + k], qacc[k]);
-  }
-  __syncthreads();
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    atomicAdd(&sum[c], csum[c]);
-    atomicAdd(&sumsq[c], csq[c]);
+    for (int k = 0; k < N; ++k) {
+      for (int off = WAVE >> 1; off >= ppr; off >>= 1) {
+        sacc[k] += __shfl_xor(sacc[k], off, WAVE);
+        qacc[k] += __shfl_xor(qacc[k], off, WAVE);
+      }
+    }
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    if (lane < ppr) {
+#pragma unroll
+      for (int k = 0; k < N; ++k) {
+        lds[wave * 2 * C + lane * N + k] = sacc[k];
+        lds[wave * 2 * C + C + lane * N + k] = qacc[k];
+      }
+    }
+    __syncthreads();
+    for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) {
+      float s = 0.0f;
+      for (int w2 = 0; w2 < nwaves; ++w2) s += lds[w2 * 2 * C + c];
+      out[c] = s;
+    }
+  } else {  // non-pow2 or very wide channel counts: LDS-atomic fallback
+    for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.0f;
+    __syncthreads();
+#pragma unroll
+    for (int k = 0; k < N; ++k) {
+      atomicAdd(&lds[cbase + k], sacc[k]);
+      atomicAdd(&lds[C + cbase + k], qacc[k]);
+    }
+    __syncthreads();
+    for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) out[c] = lds[c];
   }
 }
 
-// pass 2 (training): per-block recompute of mean/invstd from the sums
-// (C ops, cheap) -> normalize; block 0 additionally persists save_mean /
-// save_invstd for backward and EMA-updates the running stats. Folding the
-// old finalize kernel in here saves one launch per BN call (launch count is
-// what dominates small-batch BN inside graphs).
-template <typename T, bool RELU>
-__global__ void bn_norm_train_kernel(const T* __restrict__ x, T* __restrict__ y,
-                                     int64_t R, int C,
-                                     const float* __restrict__ sum,
-                                     const float* __restrict__ sumsq, float eps,
-                                     float momentum, const T* __restrict__ w,
-                                     const T* __restrict__ b,
-                                     float* __restrict__ save_mean,
-                                     float* __restrict__ save_invstd,
-                                     T* __restrict__ running_mean,
-                                     T* __restrict__ running_var) {
-  extern __shared__ float coef[];  // scale[C], shift[C]
-  float* scale = coef;
-  float* shift = coef + C;
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float m = sum[c] / (float)R;
-    float var = fmaxf(sumsq[c] / (float)R - m * m, 0.0f);
-    float inv = rsqrtf(var + eps);
-    float sc = inv * (w ? to_f(w[c]) : 1.0f);
-    scale[c] = sc;
-    shift[c] = (b ? to_f(b[c]) : 0.0f) - m * sc;
-    if (blockIdx.x == 0) {
-      save_mean[c] = m;
-      save_invstd[c] = inv;
-      if (running_mean != nullptr) {
-        from_f(running_mean[c], (1.0f - momentum) * to_f(running_mean[c]) + momentum * m);
-        float unbiased = R > 1 ? var * (float)R / (float)(R - 1) : var;
-        from_f(running_var[c], (1.0f - momentum) * to_f(running_var[c]) + momentum * unbiased);
-      }
-    }
+// finalize (fwd): reduce G partials -> mean/invstd, persist saved stats,
+// EMA-update running stats, precompute scale/shift for the norm pass.
+// Threads own channels; reads of ws[g][c] coalesce across threads.
+template <typename T>
+__global__ void bn_finalize_fwd_kernel(const float* __restrict__ ws, int G,
+                                       int64_t R, int C, float eps, float momentum,
+                                       const T* __restrict__ w,
+                                       const T* __restrict__ b,
+                                       float* __restrict__ coef,
+                                       float* __restrict__ save_mean,
+                                       float* __restrict__ save_invstd,
+                                       T* __restrict__ running_mean,
+                                       T* __restrict__ running_var) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.0f, q = 0.0f;
+  for (int g = 0; g < G; ++g) {
+    s += ws[(int64_t)g * 2 * C + c];
+    q += ws[(int64_t)g * 2 * C + C + c];
   }
+  float m = s / (float)R;
+  float var = fmaxf(q / (float)R - m * m, 0.0f);
+  float inv = rsqrtf(var + eps);
+  float sc = inv * (w ? to_f(w[c]) : 1.0f);
+  coef[c] = sc;
+  coef[C + c] = (b ? to_f(b[c]) : 0.0f) - m * sc;
+  save_mean[c] = m;
+  save_invstd[c] = inv;
+  if (running_mean != nullptr) {
+    from_f(running_mean[c], (1.0f - momentum) * to_f(running_mean[c]) + momentum * m);
+    float unbiased = R > 1 ? var * (float)R / (float)(R - 1) : var;
+    from_f(running_var[c], (1.0f - momentum) * to_f(running_var[c]) + momentum * unbiased);
+  }
+}
+
+// finalize (bwd): reduce G partials -> per-channel backward coefficients
+// g_scale = w*invstd, g_mean = sum_dy/R, g_proj = sum_dyx/R; also dweight =
+// sum_dyx, dbias = sum_dy.
+template <typename T>
+__global__ void bn_finalize_bwd_kernel(const float* __restrict__ ws, int G,
+                                       int64_t R, int C,
+                                       const float* __restrict__ invstd,
+                                       const T* __restrict__ w,
+                                       float* __restrict__ gcoef,
+                                       T* __restrict__ dweight,
+                                       T* __restrict__ dbias) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.0f, q = 0.0f;
+  for (int g = 0; g < G; ++g) {
+    s += ws[(int64_t)g * 2 * C + c];
+    q += ws[(int64_t)g * 2 * C + C + c];
+  }
+  gcoef[c] = (w ? to_f(w[c]) : 1.0f) * invstd[c];
+  gcoef[C + c] = s / (float)R;
+  gcoef[2 * C + c] = q / (float)R;
+  from_f(dweight[c], q);
+  from_f(dbias[c], s);
+}
+
+// pass 2 (training): y = [relu](scale*x + shift [+ res]); coefficients come
+// precomputed from the finalize kernel (one LDS stage per block, no
+// per-block recompute).
+template <typename T, bool RELU, bool RES>
+__global__ void bn_norm_kernel(const T* __restrict__ x, const T* __restrict__ res,
+                               T* __restrict__ y, int64_t R, int C,
+                               const float* __restrict__ coef) {
+  extern __shared__ float sc[];  // scale[C], shift[C]
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) sc[c] = coef[c];
   __syncthreads();
+  float* scale = sc;
+  float* shift = sc + C;
   constexpr int N = Pack16<T>::N;
   const int64_t nvec = R * C / N;
   const int cvec = C / N;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
     Pack16<T> pv = reinterpret_cast<const Pack16<T>*>(x)[v];
+    Pack16<T> rv;
+    if (RES) rv = reinterpret_cast<const Pack16<T>*>(res)[v];
     const int cbase = (int)(v % cvec) * N;
 #pragma unroll
     for (int k = 0; k < N; ++k) {
       float val = fmaf(to_f(pv.e[k]), scale[cbase + k], shift[cbase + k]);
+      if (RES) val += to_f(rv.e[k]);
       if (RELU) val = fmaxf(val, 0.0f);
       from_f(pv.e[k], val);
     }
@@ -685,10 +788,12 @@ __global__ void bn_norm_train_kernel(const T* __restrict__ x, T* __restrict__ y,
 
 // eval-mode forward: same as bn_norm but coefficients from running stats
 template <typename T, bool RELU>
-__global__ void bn_eval_kernel(const T* __restrict__ x, T* __restrict__ y, int64_t R,
+__global__ void bn_eval_kernel(const T* __restrict__ x, const T* __restrict__ res,
+                               T* __restrict__ y, int64_t R,
                                int C, const T* __restrict__ running_mean,
                                const T* __restrict__ running_var, float eps,
-                               const T* __restrict__ w, const T* __restrict__ b) {
+                               const T* __restrict__ w, const T* __restrict__ b,
+                               bool has_res) {
   extern __shared__ float coef[];
   float* scale = coef;
   float* shift = coef + C;
@@ -705,10 +810,13 @@ __global__ void bn_eval_kernel(const T* __restrict__ x, T* __restrict__ y, int64
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
     Pack16<T> pv = reinterpret_cast<const Pack16<T>*>(x)[v];
+    Pack16<T> rv;
+    if (has_res) rv = reinterpret_cast<const Pack16<T>*>(res)[v];
     const int cbase = (int)(v % cvec) * N;
 #pragma unroll
     for (int k = 0; k < N; ++k) {
       float val = fmaf(to_f(pv.e[k]), scale[cbase + k], shift[cbase + k]);
+      if (has_res) val += to_f(rv.e[k]);
       if (RELU) val = fmaxf(val, 0.0f);
       from_f(pv.e[k], val);
     }
@@ -716,89 +824,31 @@ __global__ void bn_eval_kernel(const T* __restrict__ x, T* __restrict__ y, int64
   }
 }
 
-// backward pass 1: per-channel sum(dy) and sum(dy * xhat), same packed
-// streaming structure as bn_sums_kernel.
-template <typename T, bool RELU>
-__global__ void bn_bwd_sums_kernel(const T* __restrict__ x, const T* __restrict__ dy,
-                                   const T* __restrict__ yout,
-                                   int64_t R, int C, const float* __restrict__ mean,
-                                   const float* __restrict__ invstd,
-                                   float* __restrict__ sum_dy,
-                                   float* __restrict__ sum_dyx) {
-  constexpr int N = Pack16<T>::N;
-  const int packs_per_row = C / N;
-  const int pk = threadIdx.x % packs_per_row;
-  const int rsub = threadIdx.x / packs_per_row;
-  const int rows_per_iter = blockDim.x / packs_per_row;
-  const int cbase = pk * N;
-
-  float m[N], is[N], sacc[N], qacc[N];
-#pragma unroll
-  for (int k = 0; k < N; ++k) {
-    m[k] = mean[cbase + k];
-    is[k] = invstd[cbase + k];
-    sacc[k] = qacc[k] = 0.0f;
-  }
-
-  if (rsub < rows_per_iter) {
-    for (int64_t r = (int64_t)blockIdx.x * rows_per_iter + rsub; r < R;
-         r += (int64_t)gridDim.x * rows_per_iter) {
-      Pack16<T> xv = *reinterpret_cast<const Pack16<T>*>(x + r * C + cbase);
-      Pack16<T> gv = *reinterpret_cast<const Pack16<T>*>(dy + r * C + cbase);
-      Pack16<T> yv;
-      if (RELU) yv = *reinterpret_cast<const Pack16<T>*>(yout + r * C + cbase);
-#pragma unroll
-      for (int k = 0; k < N; ++k) {
-        float g = to_f(gv.e[k]);
-        if (RELU && to_f(yv.e[k]) <= 0.0f) g = 0.0f;
-        float xh = (to_f(xv.e[k]) - m[k]) * is[k];
-        sacc[k] += g;
-        qacc[k] = fmaf(g, xh, qacc[k]);
-      }
-    }
-  }
-  extern __shared__ float lds[];  // csum[C], csq[C]
-  float* csum = lds;
-  float* csq = lds + C;
-  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.0f;
-  __syncthreads();
-#pragma unroll
-  for (int k = 0; k < N; ++k) {
-    atomicAdd(&csum[cbase + k], sacc[k]);
-    atomicAdd(&csq[cbase + k], qacc[k]);
-  }
-  __syncthreads();
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    atomicAdd(&sum_dy[c], csum[c]);
-    atomicAdd(&sum_dyx[c], csq[c]);
-  }
-}
-
-// backward pass 2: dx = w*invstd * (dy - sum_dy/R - xhat * sum_dyx/R)
-template <typename T, bool RELU>
+// backward pass 2: dx = g_scale * (g - g_mean - xhat * g_proj) with the
+// masked upstream grad g = dy * [yout > 0]; optionally also writes
+// dres = g (the residual branch's gradient — the ReLU follows the add, so
+// its mask applies to both branches; saves a threshold_backward + an add
+// kernel per ResNet block).
+template <typename T, bool RELU, bool DRES>
 __global__ void bn_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ dy,
                                  const T* __restrict__ yout,
-                                 T* __restrict__ dx, int64_t R, int C,
+                                 T* __restrict__ dx, T* __restrict__ dres,
+                                 int64_t R, int C,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ invstd,
-                                 const T* __restrict__ w,
-                                 const float* __restrict__ sum_dy,
-                                 const float* __restrict__ sum_dyx,
-                                 T* __restrict__ dweight, T* __restrict__ dbias) {
-  extern __shared__ float coef[];  // g1[C], g2[C], g3[C]
-  float* g_scale = coef;           // w*invstd
-  float* g_mean = coef + C;        // sum_dy / R
-  float* g_proj = coef + 2 * C;    // sum_dyx / R * invstd (per xhat term)
+                                 const float* __restrict__ gcoef) {
+  extern __shared__ float co[];  // g_scale[C], g_mean[C], g_proj[C], mean[C], invstd[C]
+  for (int c = threadIdx.x; c < 3 * C; c += blockDim.x) co[c] = gcoef[c];
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    g_scale[c] = (w ? to_f(w[c]) : 1.0f) * invstd[c];
-    g_mean[c] = sum_dy[c] / (float)R;
-    g_proj[c] = sum_dyx[c] / (float)R;
-    if (blockIdx.x == 0) {
-      from_f(dweight[c], sum_dyx[c]);
-      from_f(dbias[c], sum_dy[c]);
-    }
+    co[3 * C + c] = mean[c];
+    co[4 * C + c] = invstd[c];
   }
   __syncthreads();
+  const float* g_scale = co;
+  const float* g_mean = co + C;
+  const float* g_proj = co + 2 * C;
+  const float* mn = co + 3 * C;
+  const float* iv = co + 4 * C;
   constexpr int N = Pack16<T>::N;
   const int64_t nvec = R * C / N;
   const int cvec = C / N;
@@ -806,7 +856,7 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ 
   for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
     Pack16<T> xv = reinterpret_cast<const Pack16<T>*>(x)[v];
     Pack16<T> gv = reinterpret_cast<const Pack16<T>*>(dy)[v];
-    Pack16<T> yv;
+    Pack16<T> yv, rv;
     if (RELU) yv = reinterpret_cast<const Pack16<T>*>(yout)[v];
     const int cbase = (int)(v % cvec) * N;
 #pragma unroll
@@ -814,11 +864,13 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ 
       const int c = cbase + k;
       float g = to_f(gv.e[k]);
       if (RELU && to_f(yv.e[k]) <= 0.0f) g = 0.0f;
-      float xh = (to_f(xv.e[k]) - mean[c]) * invstd[c];
+      if (DRES) from_f(rv.e[k], g);
+      float xh = (to_f(xv.e[k]) - mn[c]) * iv[c];
       float t = g - g_mean[c] - xh * g_proj[c];
       from_f(gv.e[k], t * g_scale[c]);
     }
     reinterpret_cast<Pack16<T>*>(dx)[v] = gv;
+    if (DRES) reinterpret_cast<Pack16<T>*>(dres)[v] = rv;
   }
 }
 
@@ -1139,147 +1191,178 @@ static int64_t bn_check(const Tensor& x) {
   return x.numel() / C;
 }
 
+// grid for the partials phase: enough blocks to cover the chip, few enough
+// that the finalize reduction over G stays tiny
+static inline int bn_partials_grid(int64_t R, int rows_per_iter) {
+  static const int cap = env_int("MURMURA_BN_GRID_CAP", 256);
+  return grid_for(R, rows_per_iter, cap);
+}
+
 template <typename elem_t>
-void bn_sums_dispatch(const Tensor& x, int64_t R, int C, Tensor& sum, Tensor& sumsq) {
+static void bn_partials_launch(const Tensor& x, const Tensor* dy, const Tensor* yout,
+                               int64_t R, int C, const float* mean,
+                               const float* invstd, Tensor& ws, int G, bool relu) {
   constexpr int N = Pack16<elem_t>::N;
-  const int packs_per_row = C / N;
-  const int rows_per_iter = std::max(1, BLOCK / packs_per_row);
-  // cap the grid low: the cross-block combine is atomics onto only 2*C
-  // addresses, so 2048 blocks serialize ~2048 adds per address (measured
-  // 31 us for an 8 MB reduction); 256 blocks iterate instead and contend 8x
-  // less while still covering all 256 CUs
-  static const int cap = env_int("MURMURA_BN_GRID_CAP", 64);
-  int blocks = grid_for(R, rows_per_iter, cap);
-  size_t lds = 2 * C * sizeof(float);
-  bn_sums_kernel<elem_t><<<blocks, BLOCK, lds, cur_stream()>>>(
-      (const elem_t*)x.data_ptr(), R, C, sum.data_ptr<float>(), sumsq.data_ptr<float>());
+  const int nwaves = BLOCK / WAVE;
+  const int ppr = C / N;
+  const bool pow2 = (ppr & (ppr - 1)) == 0 && ppr <= WAVE;
+  size_t lds = (pow2 ? nwaves * 2 * C : 2 * C) * sizeof(float);
+  if (dy == nullptr) {
+    bn_partials_kernel<elem_t, false, false><<<G, BLOCK, lds, cur_stream()>>>(
+        (const elem_t*)x.data_ptr(), nullptr, nullptr, R, C, nullptr, nullptr,
+        ws.data_ptr<float>());
+  } else if (relu) {
+    bn_partials_kernel<elem_t, true, true><<<G, BLOCK, lds, cur_stream()>>>(
+        (const elem_t*)x.data_ptr(), (const elem_t*)dy->data_ptr(),
+        (const elem_t*)yout->data_ptr(), R, C, mean, invstd, ws.data_ptr<float>());
+  } else {
+    bn_partials_kernel<elem_t, true, false><<<G, BLOCK, lds, cur_stream()>>>(
+        (const elem_t*)x.data_ptr(), (const elem_t*)dy->data_ptr(), nullptr, R, C,
+        mean, invstd, ws.data_ptr<float>());
+  }
 }
 
 std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optional<Tensor> b,
                                  c10::optional<Tensor> running_mean,
                                  c10::optional<Tensor> running_var, double momentum,
-                                 double eps, bool relu) {
+                                 double eps, bool relu, c10::optional<Tensor> res) {
   int64_t R = bn_check(x);
   int C = (int)x.size(1);
   auto fopt = x.options().dtype(at::kFloat);
-  Tensor ws = at::zeros({2, C}, fopt);     // one fill covers sum + sumsq
-  Tensor sum = ws[0], sumsq = ws[1];
-  Tensor saved = at::empty({2, C}, fopt);  // save_mean + save_invstd
-  DISPATCH_FT(x, { bn_sums_dispatch<elem_t>(x, R, C, sum, sumsq); });
   if (running_mean.has_value()) {
     TORCH_CHECK(running_mean->scalar_type() == x.scalar_type(),
                 "bn: running stats must match input dtype");
   }
+  if (res.has_value()) {
+    TORCH_CHECK(res->is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                    res->sizes() == x.sizes() && res->scalar_type() == x.scalar_type(),
+                "bn: residual must be channels_last, same shape/dtype");
+  }
+  Tensor saved = at::empty({2, C}, fopt);  // save_mean + save_invstd
+  Tensor coef = at::empty({2 * C}, fopt);  // scale + shift
   Tensor y = at::empty_like(x);
-  int blocks = grid_for(R * C / 4, BLOCK);
-  size_t lds = 2 * C * sizeof(float);
   float* sm = saved.data_ptr<float>();
   DISPATCH_FT(x, {
+    constexpr int N = Pack16<elem_t>::N;
+    const int rows_per_iter = std::max(1, BLOCK / (C / N));
+    const int G = bn_partials_grid(R, rows_per_iter);
+    Tensor ws = at::empty({G, 2 * C}, fopt);  // partials: no pre-zeroing
+    bn_partials_launch<elem_t>(x, nullptr, nullptr, R, C, nullptr, nullptr, ws, G,
+                               false);
     elem_t* rm = running_mean.has_value() ? (elem_t*)running_mean->data_ptr() : nullptr;
     elem_t* rv = running_var.has_value() ? (elem_t*)running_var->data_ptr() : nullptr;
-    if (relu) {
-      bn_norm_train_kernel<elem_t, true><<<blocks, BLOCK, lds, cur_stream()>>>(
-          (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
-          sum.data_ptr<float>(), sumsq.data_ptr<float>(), (float)eps, (float)momentum,
-          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr,
-          sm, sm + C, rm, rv);
+    bn_finalize_fwd_kernel<elem_t>
+        <<<grid_for(C, BLOCK), BLOCK, 0, cur_stream()>>>(
+            ws.data_ptr<float>(), G, R, C, (float)eps, (float)momentum,
+            w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+            b.has_value() ? (const elem_t*)b->data_ptr() : nullptr,
+            coef.data_ptr<float>(), sm, sm + C, rm, rv);
+    int blocks = grid_for(R * C / N, BLOCK);
+    size_t lds = 2 * C * sizeof(float);
+    const elem_t* rp = res.has_value() ? (const elem_t*)res->data_ptr() : nullptr;
+    if (relu && rp) {
+      bn_norm_kernel<elem_t, true, true><<<blocks, BLOCK, lds, cur_stream()>>>(
+          (const elem_t*)x.data_ptr(), rp, (elem_t*)y.data_ptr(), R, C,
+          coef.data_ptr<float>());
+    } else if (relu) {
+      bn_norm_kernel<elem_t, true, false><<<blocks, BLOCK, lds, cur_stream()>>>(
+          (const elem_t*)x.data_ptr(), nullptr, (elem_t*)y.data_ptr(), R, C,
+          coef.data_ptr<float>());
+    } else if (rp) {
+      bn_norm_kernel<elem_t, false, true><<<blocks, BLOCK, lds, cur_stream()>>>(
+          (const elem_t*)x.data_ptr(), rp, (elem_t*)y.data_ptr(), R, C,
+          coef.data_ptr<float>());
     } else {
-      bn_norm_train_kernel<elem_t, false><<<blocks, BLOCK, lds, cur_stream()>>>(
-          (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
-          sum.data_ptr<float>(), sumsq.data_ptr<float>(), (float)eps, (float)momentum,
-          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr,
-          sm, sm + C, rm, rv);
+      bn_norm_kernel<elem_t, false, false><<<blocks, BLOCK, lds, cur_stream()>>>(
+          (const elem_t*)x.data_ptr(), nullptr, (elem_t*)y.data_ptr(), R, C,
+          coef.data_ptr<float>());
     }
   });
   return {y, saved[0], saved[1]};
 }
 
 Tensor bn_fwd_eval(Tensor x, c10::optional<Tensor> w, c10::optional<Tensor> b,
-                   Tensor running_mean, Tensor running_var, double eps, bool relu) {
+                   Tensor running_mean, Tensor running_var, double eps, bool relu,
+                   c10::optional<Tensor> res) {
   int64_t R = bn_check(x);
   int C = (int)x.size(1);
   Tensor y = at::empty_like(x);
-  int blocks = grid_for(R * C / 4, BLOCK);
   size_t lds = 2 * C * sizeof(float);
   TORCH_CHECK(running_mean.scalar_type() == x.scalar_type(),
               "bn: running stats must match input dtype");
   DISPATCH_FT(x, {
+    constexpr int N = Pack16<elem_t>::N;
+    int blocks = grid_for(R * C / N, BLOCK);
+    const elem_t* rp = res.has_value() ? (const elem_t*)res->data_ptr() : nullptr;
     if (relu) {
       bn_eval_kernel<elem_t, true><<<blocks, BLOCK, lds, cur_stream()>>>(
-          (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
+          (const elem_t*)x.data_ptr(), rp, (elem_t*)y.data_ptr(), R, C,
           (const elem_t*)running_mean.data_ptr(),
           (const elem_t*)running_var.data_ptr(), (float)eps,
           w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr);
+          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr, rp != nullptr);
     } else {
       bn_eval_kernel<elem_t, false><<<blocks, BLOCK, lds, cur_stream()>>>(
-          (const elem_t*)x.data_ptr(), (elem_t*)y.data_ptr(), R, C,
+          (const elem_t*)x.data_ptr(), rp, (elem_t*)y.data_ptr(), R, C,
           (const elem_t*)running_mean.data_ptr(),
           (const elem_t*)running_var.data_ptr(), (float)eps,
           w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr);
+          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr, rp != nullptr);
     }
   });
   return y;
 }
 
 std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor mean,
-                           Tensor invstd, c10::optional<Tensor> yout, bool relu) {
+                           Tensor invstd, c10::optional<Tensor> yout, bool relu,
+                           bool need_dres) {
   TORCH_CHECK(!relu || yout.has_value(), "bn_bwd: relu needs the saved output");
+  TORCH_CHECK(!need_dres || relu, "bn_bwd: residual fusion implies fused relu");
   int64_t R = bn_check(x);
   TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast));
   int C = (int)x.size(1);
   auto fopt = x.options().dtype(at::kFloat);
-  Tensor ws = at::zeros({2, C}, fopt);
-  Tensor sum_dy = ws[0], sum_dyx = ws[1];
+  Tensor gcoef = at::empty({3 * C}, fopt);
+  Tensor dx = at::empty_like(dy);
+  Tensor dres = need_dres ? at::empty_like(dy) : Tensor();
+  Tensor dweight = at::empty({C}, x.options());
+  Tensor dbias = at::empty({C}, x.options());
   hipStream_t st = cur_stream();
   DISPATCH_FT(x, {
     constexpr int N = Pack16<elem_t>::N;
-    const int packs_per_row = C / N;
-    const int rows_per_iter = std::max(1, BLOCK / packs_per_row);
-    static const int cap2 = env_int("MURMURA_BN_GRID_CAP", 64);
-    int blocks = grid_for(R, rows_per_iter, cap2);
-    size_t lds = 2 * C * sizeof(float);
-    const elem_t* yp = yout.has_value() ? (const elem_t*)yout->data_ptr() : nullptr;
-    if (relu) {
-      bn_bwd_sums_kernel<elem_t, true><<<blocks, BLOCK, lds, st>>>(
-          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp, R, C,
-          mean.data_ptr<float>(), invstd.data_ptr<float>(),
-          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
+    const int rows_per_iter = std::max(1, BLOCK / (C / N));
+    const int G = bn_partials_grid(R, rows_per_iter);
+    Tensor ws = at::empty({G, 2 * C}, fopt);  // partials: no pre-zeroing
+    const Tensor* yp = yout.has_value() ? &*yout : nullptr;
+    bn_partials_launch<elem_t>(x, &dy, yp, R, C, mean.data_ptr<float>(),
+                               invstd.data_ptr<float>(), ws, G, relu);
+    bn_finalize_bwd_kernel<elem_t>
+        <<<grid_for(C, BLOCK), BLOCK, 0, st>>>(
+            ws.data_ptr<float>(), G, R, C, invstd.data_ptr<float>(),
+            w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+            gcoef.data_ptr<float>(), (elem_t*)dweight.data_ptr(),
+            (elem_t*)dbias.data_ptr());
+    int blocks = grid_for(R * C / N, BLOCK);
+    size_t lds = 5 * C * sizeof(float);
+    const elem_t* yp2 = yp ? (const elem_t*)yp->data_ptr() : nullptr;
+    if (relu && need_dres) {
+      bn_bwd_dx_kernel<elem_t, true, true><<<blocks, BLOCK, lds, st>>>(
+          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp2,
+          (elem_t*)dx.data_ptr(), (elem_t*)dres.data_ptr(), R, C,
+          mean.data_ptr<float>(), invstd.data_ptr<float>(), gcoef.data_ptr<float>());
+    } else if (relu) {
+      bn_bwd_dx_kernel<elem_t, true, false><<<blocks, BLOCK, lds, st>>>(
+          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp2,
+          (elem_t*)dx.data_ptr(), nullptr, R, C,
+          mean.data_ptr<float>(), invstd.data_ptr<float>(), gcoef.data_ptr<float>());
     } else {
-      bn_bwd_sums_kernel<elem_t, false><<<blocks, BLOCK, lds, st>>>(
-          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp, R, C,
-          mean.data_ptr<float>(), invstd.data_ptr<float>(),
-          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>());
+      bn_bwd_dx_kernel<elem_t, false, false><<<blocks, BLOCK, lds, st>>>(
+          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp2,
+          (elem_t*)dx.data_ptr(), nullptr, R, C,
+          mean.data_ptr<float>(), invstd.data_ptr<float>(), gcoef.data_ptr<float>());
     }
   });
-  Tensor dx = at::empty_like(dy);
-  Tensor dweight = at::empty({C}, x.options());
-  Tensor dbias = at::empty({C}, x.options());
-  int blocks = grid_for(R * C / 4, BLOCK);
-  size_t lds = 3 * C * sizeof(float);
-  DISPATCH_FT(x, {
-    const elem_t* yp2 = yout.has_value() ? (const elem_t*)yout->data_ptr() : nullptr;
-    if (relu) {
-      bn_bwd_dx_kernel<elem_t, true><<<blocks, BLOCK, lds, st>>>(
-          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp2,
-          (elem_t*)dx.data_ptr(), R, C, mean.data_ptr<float>(),
-          invstd.data_ptr<float>(),
-          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
-          (elem_t*)dweight.data_ptr(), (elem_t*)dbias.data_ptr());
-    } else {
-      bn_bwd_dx_kernel<elem_t, false><<<blocks, BLOCK, lds, st>>>(
-          (const elem_t*)x.data_ptr(), (const elem_t*)dy.data_ptr(), yp2,
-          (elem_t*)dx.data_ptr(), R, C, mean.data_ptr<float>(),
-          invstd.data_ptr<float>(),
-          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-          sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
-          (elem_t*)dweight.data_ptr(), (elem_t*)dbias.data_ptr());
-    }
-  });
+  if (need_dres) return {dx, dweight, dbias, dres};
   return {dx, dweight, dbias};
 }
 
@@ -1287,16 +1370,22 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train,
-        "K14: fused NHWC BatchNorm(+ReLU) training forward -> (y, mean, invstd)",
+        "K14: fused NHWC BatchNorm(+ReLU)(+residual) training forward -> "
+        "(y, mean, invstd)",
         py::arg("x"), py::arg("w"), py::arg("b"), py::arg("running_mean"),
         py::arg("running_var"), py::arg("momentum"), py::arg("eps"),
-        py::arg("relu") = false);
-  m.def("bn_fwd_eval", &bn_fwd_eval, "K14: NHWC BatchNorm(+ReLU) eval forward",
+        py::arg("relu") = false, py::arg("res") = py::none());
+  m.def("bn_fwd_eval", &bn_fwd_eval,
+        "K14: NHWC BatchNorm(+ReLU)(+residual) eval forward",
         py::arg("x"), py::arg("w"), py::arg("b"), py::arg("running_mean"),
-        py::arg("running_var"), py::arg("eps"), py::arg("relu") = false);
-  m.def("bn_bwd", &bn_bwd, "K14: NHWC BatchNorm(+ReLU) backward -> (dx, dweight, dbias)",
+        py::arg("running_var"), py::arg("eps"), py::arg("relu") = false,
+        py::arg("res") = py::none());
+  m.def("bn_bwd", &bn_bwd,
+        "K14: NHWC BatchNorm(+ReLU)(+residual) backward -> "
+        "(dx, dweight, dbias[, dres])",
         py::arg("x"), py::arg("dy"), py::arg("w"), py::arg("mean"), py::arg("invstd"),
-        py::arg("yout") = py::none(), py::arg("relu") = false);
+        py::arg("yout") = py::none(), py::arg("relu") = false,
+        py::arg("need_dres") = false);
   m.def("weighted_sum", &weighted_sum, "K1: out = sum_i w_i * x_i",
         py::arg("stacked"), py::arg("w"), py::arg("out") = py::none());
   m.def("pairwise_sq_dists", &pairwise_sq_dists, "K2: [m,m] squared L2 matrix");

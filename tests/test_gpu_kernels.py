@@ -283,3 +283,71 @@ def test_fused_bn_relu_matches_torch(dtype):
         ye = fused(x.contiguous(memory_format=torch.channels_last))
         yre = torch.relu(refbn(x.float()))
     assert torch.allclose(ye.float(), yre, atol=tol, rtol=tol)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("c", [64, 256])
+def test_fused_bn_add_relu_matches_torch(dtype, c):
+    """Residual-fused tail: y = relu(bn(x) + res) forward AND both input
+    gradients from one backward launch (the ResNet BasicBlock tail)."""
+    from murmura_amd.ops.fused_bn import MurmuraBNAddReLU
+
+    torch.manual_seed(c)
+    n, h, w = 8, 8, 8
+    x = torch.randn(n, c, h, w, device="cuda", dtype=dtype)
+    r = torch.randn(n, c, h, w, device="cuda", dtype=dtype)
+    xf = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    rf = r.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    xr = x.clone().float().requires_grad_(True)
+    rr = r.clone().float().requires_grad_(True)
+
+    fused = MurmuraBNAddReLU(c).cuda().to(dtype)
+    refbn = nn.BatchNorm2d(c).cuda().float()
+    refbn.load_state_dict({k: v.float() for k, v in fused.state_dict().items()})
+    fused.train(); refbn.train()
+
+    y = fused(xf, res=rf)
+    yr = torch.relu(refbn(xr) + rr)
+    tol = 1e-5 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(y.float(), yr, atol=tol, rtol=tol)
+
+    g = torch.randn_like(yr)
+    y.backward(g.to(dtype).contiguous(memory_format=torch.channels_last))
+    yr.backward(g)
+    gtol = 1e-4 if dtype == torch.float32 else 6e-2
+    for got, ref in [(xf.grad, xr.grad), (rf.grad, rr.grad)]:
+        scale = ref.abs().max().clamp_min(1e-6)
+        assert ((got.float() - ref).abs().max() / scale).item() < gtol
+    wscale = refbn.weight.grad.abs().max().clamp_min(1e-6)
+    assert ((fused.weight.grad.float() - refbn.weight.grad).abs().max() / wscale).item() < gtol
+
+    # eval mode with residual
+    fused.eval(); refbn.eval()
+    with torch.no_grad():
+        ye = fused(x.contiguous(memory_format=torch.channels_last),
+                   res=r.contiguous(memory_format=torch.channels_last))
+        yre = torch.relu(refbn(x.float()) + r.float())
+    assert torch.allclose(ye.float(), yre, atol=tol, rtol=tol)
+
+
+def test_fused_bn_add_relu_cpu_fallback():
+    from murmura_amd.ops.fused_bn import MurmuraBNAddReLU
+
+    m = MurmuraBNAddReLU(16)
+    r = nn.BatchNorm2d(16)
+    r.load_state_dict(m.state_dict())
+    x = torch.randn(4, 16, 5, 5)
+    res = torch.randn(4, 16, 5, 5)
+    assert torch.allclose(m(x, res=res), torch.relu(r(x) + res))
+
+
+def test_deferred_num_batches_tracked_bump():
+    from murmura_amd.ops.fused_bn import (
+        MurmuraBatchNorm2d,
+        bump_num_batches_tracked,
+    )
+
+    m = nn.Sequential(MurmuraBatchNorm2d(8), MurmuraBatchNorm2d(8))
+    bump_num_batches_tracked(m, 5)
+    assert m[0].num_batches_tracked.item() == 5
+    assert m[1].num_batches_tracked.item() == 5

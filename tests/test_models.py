@@ -99,3 +99,32 @@ def test_correct_prediction_lowers_evidential_loss():
     right = torch.tensor([[5.0, -5.0, -5.0]])
     wrong = torch.tensor([[-5.0, 5.0, -5.0]])
     assert loss_fn(right, y, 0) < loss_fn(wrong, y, 0)
+
+
+def test_basic_block_residual_fusion_cpu_semantics():
+    """BasicBlock's fused tail relu(bn2(conv2)+shortcut) must equal the plain
+    composition on the CPU fallback path (the GPU kernel is compared to torch
+    in tests/test_gpu_kernels.py)."""
+    import torch
+    from torch import nn
+
+    from murmura_amd.models.zoo import BasicBlock
+
+    torch.manual_seed(0)
+    blk = BasicBlock(8, 8)
+    blk.train()
+    x = torch.randn(4, 8, 6, 6)
+    out = blk(x)
+
+    # reference composition using the same parameters
+    ref_bn2 = nn.BatchNorm2d(8)
+    ref_bn2.load_state_dict(
+        {k.split("bn2.")[1]: v for k, v in blk.state_dict().items()
+         if k.startswith("bn2.")}
+    )
+    torch.manual_seed(0)
+    blk2 = BasicBlock(8, 8)
+    blk2.load_state_dict(blk.state_dict())
+    h = blk2.bn1(blk2.conv1(x))
+    expected = torch.relu(ref_bn2(blk2.conv2(h)) + x)
+    assert torch.allclose(out, expected, atol=1e-5)
