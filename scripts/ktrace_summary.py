@@ -27,7 +27,15 @@ def main():
     t0 = min(r[1] for r in rows)
     t1 = max(r[2] for r in rows)
     print(f"total window {(t1 - t0) / 1e9:.2f}s, {len(rows)} dispatches")
-    lo = t0 + int((t1 - t0) * (1 - frac))
+    # anchor the steady window to the sgd_step dispatches (one per train
+    # batch, absent from capture warmup / MIOpen find): last `frac` of them
+    sgd = sorted(r[1] for r in rows if "sgd_step" in r[0])
+    if sgd:
+        n = max(1, int(len(sgd) * frac))
+        lo = sgd[-n]
+        print(f"anchor: last {n} of {len(sgd)} sgd steps")
+    else:
+        lo = t0 + int((t1 - t0) * (1 - frac))
     ss = [r for r in rows if r[1] >= lo]
     sswall = (t1 - lo) / 1e9
     agg = defaultdict(lambda: [0, 0])
