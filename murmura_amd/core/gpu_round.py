@@ -82,32 +82,39 @@ class TrainGraph:
             self.static_x = self.static_x.contiguous(memory_format=torch.channels_last)
         self.static_y = torch.zeros((n_used,), device=shard.device, dtype=shard.y.dtype)
         self.loss_sum = torch.zeros((), device=shard.device, dtype=torch.float32)
-        self.lr = torch.zeros((), device=shard.device, dtype=torch.float32)
         # device-resident KL annealing weight for EvidentialLoss capture
         self.kl_weight = torch.zeros((), device=shard.device, dtype=torch.float32)
         self.graph: Optional[torch.cuda.CUDAGraph] = None
+        self._params = [p for p in node.model.parameters() if p.requires_grad]
+        self._captured_lr: Optional[float] = None
         self._gen = torch.Generator().manual_seed(
             0x5EED ^ (node.node_id * 0x9E3779B9 & 0x7FFFFFFF)
         )
 
     def _one_step(self, s: int) -> None:
+        """One training step inside the capture.
+
+        Gradients flow through ``torch.autograd.grad`` (NOT ``backward``):
+        ``backward`` into pre-existing flat-view ``.grad`` buffers records one
+        AccumulateGrad ``add_`` kernel PER PARAMETER per step (62 launches per
+        ResNet-18 batch — ~300 us of pure kernel-floor cost, measured), while
+        ``grad()`` routes each producer kernel's output directly. The SGD
+        update is one horizontally-fused ``_foreach_add_`` family over the
+        param views (lr is baked into the capture; run_epoch recaptures when
+        lr changes — FL configs train at constant lr)."""
         node = self.node
         x = self.static_x[s * self.bs : (s + 1) * self.bs]
         y = self.static_y[s * self.bs : (s + 1) * self.bs]
         from murmura_amd.models.evidential import EvidentialLoss
 
-        node.store.zero_grad()
         out = node.model(x)
         if isinstance(node.criterion, EvidentialLoss):
             loss = node.criterion(out, y, kl_weight=self.kl_weight)
         else:
             loss = torch.nn.functional.cross_entropy(out.float(), y)
-        loss.backward()
-        # fused SGD with a DEVICE-tensor lr so replays honor lr changes
-        from murmura_amd import ops as _ops
-
-        pflat = node.store.flat[: node.store.spec.param_numel]
-        _ops.sgd_step_lrt(pflat, node.store.grad_flat, self.lr)
+        params = self._params
+        grads = torch.autograd.grad(loss, params)
+        torch._foreach_add_(params, grads, alpha=-self._captured_lr)
         self.loss_sum += loss.detach()
 
     def _capture(self) -> None:
@@ -115,7 +122,6 @@ class TrainGraph:
 
         node = self.node
         node.model.train()
-        node.store.ensure_grads()
         # warmup on a side stream executes real steps — snapshot/restore the
         # flat state so capture-time warmup does not perturb training
         saved = node.store.flat.clone()
@@ -150,10 +156,12 @@ class TrainGraph:
         # fill static buffers BEFORE the first capture: warmup executes real
         # kernels and must see genuine inputs/targets
         self.shard.shuffled(self.static_x, self.static_y, self._gen)
-        self.lr.fill_(lr)
         if isinstance(self.node.criterion, EvidentialLoss):
             self.kl_weight.fill_(self.node.criterion.kl_weight_at(round_num))
+        if self._captured_lr is not None and lr != self._captured_lr:
+            self.graph = None  # lr is baked into the capture; recapture
         if self.graph is None:
+            self._captured_lr = float(lr)
             self._capture()
         self.graph.replay()
         return self.loss_sum
