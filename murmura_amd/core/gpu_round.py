@@ -114,7 +114,8 @@ class TrainGraph:
             loss = torch.nn.functional.cross_entropy(out.float(), y)
         params = self._params
         grads = torch.autograd.grad(loss, params)
-        torch._foreach_add_(params, grads, alpha=-self._captured_lr)
+        with torch.no_grad():
+            torch._foreach_add_(params, grads, alpha=-self._captured_lr)
         self.loss_sum += loss.detach()
 
     def _capture(self) -> None:
