@@ -114,10 +114,26 @@ def krum_select(d2: Tensor, num_compromised: int) -> Tensor:
 
 
 # ------------------------------------------------------------------ K4/K5
+def _packed_sketch_table(hash_idx: Tensor, signs: Tensor, device) -> Tensor:
+    """hash+sign packed into one int32 per element (bin in the low 31 bits,
+    sign in the sign bit) — halves the kernel's table traffic. Cached on the
+    hash tensor (the aggregator reuses its seeded tables every round)."""
+    packed = getattr(hash_idx, "_murmura_packed", None)
+    if packed is None or packed.device != device:
+        neg = (signs < 0).to(torch.int32)
+        packed = (hash_idx.to(torch.int32) | (neg << 31)).to(device).contiguous()
+        try:
+            hash_idx._murmura_packed = packed
+        except Exception:
+            pass
+    return packed
+
+
 def count_sketch(stacked: Tensor, hash_idx: Tensor, signs: Tensor, sketch_size: int) -> Tensor:
     if _use_native(stacked if stacked.dim() == 2 else stacked.view(1, -1)):
         x = stacked if stacked.dim() == 2 else stacked.view(1, -1)
-        out = _EXT.count_sketch(x, hash_idx, signs, sketch_size)
+        pt = _packed_sketch_table(hash_idx, signs, x.device)
+        out = _EXT.count_sketch(x, pt, sketch_size)
         return out if stacked.dim() == 2 else out.view(-1)
     return ref.count_sketch(stacked, hash_idx, signs, sketch_size)
 

@@ -74,10 +74,15 @@ def main():
         del x, g, p
         torch.cuda.empty_cache()
 
-    # m=16 gram (register-pressure case)
+    # m=11/16 gram (register-pressure cases; m=11 = 10-node reference recipe)
     x = torch.randn(16, 20_000_000, device=dev, dtype=torch.float32)
     t = timeit(lambda: ops.pairwise_sq_dists(x))
     report("K2 pairwise m=16 P=20M fp32", t, 16 * 20_000_000 * 4)
+    x11 = x[:11]
+    t = timeit(lambda: ops.pairwise_sq_dists(x11))
+    report("K2 pairwise m=11 P=20M fp32", t, 11 * 20_000_000 * 4)
+    del x, x11
+    torch.cuda.empty_cache()
 
     # eval epilogues
     logits = torch.randn(4096, 62, device=dev, dtype=torch.bfloat16)
