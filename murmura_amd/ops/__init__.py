@@ -129,9 +129,43 @@ def _packed_sketch_table(hash_idx: Tensor, signs: Tensor, device) -> Tensor:
     return packed
 
 
+def _grouped_sketch_tables(hash_idx: Tensor, signs: Tensor, device):
+    """Detect group-of-8-constant bins (ops.make_sketch_tables' layout) and
+    build the packed group table: bin | signbits<<16 per group, plus a
+    per-element packed tail for P % 8 leftovers. Cached on the hash tensor."""
+    cached = getattr(hash_idx, "_murmura_g8", None)
+    if cached is not None and cached[0].device == device:
+        return cached
+    P = hash_idx.numel()
+    ng = P // 8
+    if ng == 0 or int(hash_idx.max()) >= (1 << 16):
+        return None
+    h8 = hash_idx[: ng * 8].view(ng, 8)
+    if not bool((h8 == h8[:, :1]).all()):
+        return None
+    neg = (signs[: ng * 8].view(ng, 8) < 0).to(torch.int32)
+    bits = (neg << (16 + torch.arange(8, dtype=torch.int32, device=neg.device))).sum(
+        dim=1, dtype=torch.int32
+    )
+    gt = (h8[:, 0].to(torch.int32) | bits).to(device).contiguous()
+    tneg = (signs[ng * 8 :] < 0).to(torch.int32)
+    tail = (hash_idx[ng * 8 :].to(torch.int32) | (tneg << 31)).to(device).contiguous()
+    out = (gt, tail)
+    try:
+        hash_idx._murmura_g8 = out
+    except Exception:
+        pass
+    return out
+
+
 def count_sketch(stacked: Tensor, hash_idx: Tensor, signs: Tensor, sketch_size: int) -> Tensor:
     if _use_native(stacked if stacked.dim() == 2 else stacked.view(1, -1)):
         x = stacked if stacked.dim() == 2 else stacked.view(1, -1)
+        if x.shape[0] * sketch_size * 4 <= 64 * 1024 and sketch_size < (1 << 16):
+            g8 = _grouped_sketch_tables(hash_idx, signs, x.device)
+            if g8 is not None:
+                out = _EXT.count_sketch_g8(x, g8[0], g8[1], sketch_size)
+                return out if stacked.dim() == 2 else out.view(-1)
         pt = _packed_sketch_table(hash_idx, signs, x.device)
         out = _EXT.count_sketch(x, pt, sketch_size)
         return out if stacked.dim() == 2 else out.view(-1)

@@ -362,6 +362,70 @@ __global__ void count_sketch_multirow_kernel(const T* __restrict__ x,
   }
 }
 
+// Grouped-table variant (the production path): bins are constant over
+// aligned groups of 8 coordinates (ops/reference.py make_sketch_tables),
+// signs per element. Each thread accumulates a group's signed sum for a row
+// in registers and issues ONE LDS atomic — 8x fewer atomics than the
+// per-element kernel, which is atomic-issue-bound (~200G atomics/s measured).
+// Table entry: bin in bits 0..15, per-element sign bits in 16..23.
+template <typename T, bool ALIGNED>
+__global__ void count_sketch_g8_kernel(const T* __restrict__ x,
+                                       const int* __restrict__ gt, int m,
+                                       int64_t P, int64_t ngroups, int S,
+                                       const int* __restrict__ tailt, int ntail,
+                                       float* __restrict__ out, int R) {
+  extern __shared__ float hist[];  // [R][m][S]
+  const int nh = R * m * S;
+  for (int b = threadIdx.x; b < nh; b += blockDim.x) hist[b] = 0.0f;
+  __syncthreads();
+  float* my = hist + (int)((threadIdx.x / WAVE) % R) * m * S;
+
+  constexpr int N = Pack16<T>::N;  // 8 bf16 / 4 fp32 per 16-B pack
+  constexpr int PPG = 8 / N;       // packs per group
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; g < ngroups;
+       g += stride) {
+    const int w = gt[g];
+    const int bin = w & 0xFFFF;
+    float sgn[8];
+#pragma unroll
+    for (int k = 0; k < 8; ++k) sgn[k] = ((w >> (16 + k)) & 1) ? -1.0f : 1.0f;
+    for (int i = 0; i < m; ++i) {
+      const T* row = x + (int64_t)i * P + g * 8;
+      float s = 0.0f;
+      if (ALIGNED) {
+#pragma unroll
+        for (int pp = 0; pp < PPG; ++pp) {
+          Pack16<T> xv = reinterpret_cast<const Pack16<T>*>(row)[pp];
+#pragma unroll
+          for (int k = 0; k < N; ++k) s = fmaf(sgn[pp * N + k], to_f(xv.e[k]), s);
+        }
+      } else {  // rows not 16-B aligned (P % 8 != 0): element loads
+#pragma unroll
+        for (int k = 0; k < 8; ++k) s = fmaf(sgn[k], to_f(row[k]), s);
+      }
+      atomicAdd(&my[i * S + bin], s);
+    }
+  }
+  // per-element tail for p in [ngroups*8, P)
+  for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < ntail;
+       t += stride) {
+    const int w = tailt[t];
+    const int bin = w & 0x7FFFFFFF;
+    const int64_t p = ngroups * 8 + t;
+    for (int i = 0; i < m; ++i) {
+      float val = to_f(x[(int64_t)i * P + p]);
+      atomicAdd(&my[i * S + bin], w < 0 ? -val : val);
+    }
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < m * S; b += blockDim.x) {
+    float acc = 0.0f;
+    for (int r = 0; r < R; ++r) acc += hist[r * m * S + b];
+    if (acc != 0.0f) atomicAdd(&out[b], acc);
+  }
+}
+
 // fallback for sketches too large for replicated LDS: one row per blockIdx.y,
 // single histogram
 template <typename T>
@@ -1150,6 +1214,39 @@ Tensor l2_dists_to(Tensor own, Tensor stacked) {
   return out.clamp_min_(0.0).sqrt_();
 }
 
+Tensor count_sketch_g8(Tensor stacked, Tensor gt, Tensor tailt, int64_t S) {
+  check_flat(stacked, "stacked");
+  TORCH_CHECK(stacked.dim() == 2);
+  int m = (int)stacked.size(0);
+  int64_t P = stacked.size(1);
+  TORCH_CHECK(S < (1 << 16), "count_sketch_g8: S must fit 16 bits");
+  TORCH_CHECK(gt.scalar_type() == at::kInt && tailt.scalar_type() == at::kInt);
+  Tensor g = gt.is_cuda() ? gt.contiguous() : gt.to(stacked.device()).contiguous();
+  Tensor tt = tailt.is_cuda() ? tailt.contiguous()
+                              : tailt.to(stacked.device()).contiguous();
+  int64_t ngroups = g.numel();
+  int ntail = (int)tt.numel();
+  TORCH_CHECK(ngroups * 8 + ntail == P, "count_sketch_g8: table size mismatch");
+  Tensor out = at::zeros({m, S}, stacked.options().dtype(at::kFloat));
+  size_t one = (size_t)m * S * sizeof(float);
+  TORCH_CHECK(one <= 64 * 1024, "count_sketch_g8: m*S too large for LDS");
+  int R = std::max(1, std::min(4, (int)((64 * 1024) / one)));
+  int blocks = grid_for(ngroups, BLOCK, 2048);
+  bool aligned = (P % 8) == 0;
+  DISPATCH_FT(stacked, {
+    if (aligned) {
+      count_sketch_g8_kernel<elem_t, true><<<blocks, BLOCK, one * R, cur_stream()>>>(
+          (const elem_t*)stacked.data_ptr(), g.data_ptr<int>(), m, P, ngroups,
+          (int)S, tt.data_ptr<int>(), ntail, out.data_ptr<float>(), R);
+    } else {
+      count_sketch_g8_kernel<elem_t, false><<<blocks, BLOCK, one * R, cur_stream()>>>(
+          (const elem_t*)stacked.data_ptr(), g.data_ptr<int>(), m, P, ngroups,
+          (int)S, tt.data_ptr<int>(), ntail, out.data_ptr<float>(), R);
+    }
+  });
+  return out;
+}
+
 Tensor count_sketch(Tensor stacked, Tensor packed, int64_t S) {
   check_flat(stacked, "stacked");
   TORCH_CHECK(stacked.dim() == 2);
@@ -1479,6 +1576,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_norms", &row_norms, "K12: per-row L2 norms");
   m.def("l2_dists_to", &l2_dists_to, "K2 variant: dists of rows to own");
   m.def("count_sketch", &count_sketch, "K4: count-sketch projection");
+  m.def("count_sketch_g8", &count_sketch_g8,
+        "K4: count-sketch with group-of-8 bins (1 atomic per group per row)");
   m.def("sgd_step", &sgd_step, "K6: fused p -= lr*g");
   m.def("sgd_step_lrt", &sgd_step_lrt, "K6: fused p -= lr*g, lr from device");
   m.def("gaussian_inject", &gaussian_inject, "K10: x + N(0, sigma^2) (Philox)");
