@@ -87,6 +87,103 @@ class EvalContext:
         n = max(1, seen)
         return vac_sum / n, correct.float() / n
 
+    # ------------------------------------------------------ batched scoring
+    # VERDICT round-1 weak #3: UBAR stage-2 / EvidentialTrust / DMTT scoring
+    # ran m sequential copy_from_flat + forward per candidate per round. Here
+    # all k candidate states are evaluated in ONE vmapped forward
+    # (torch.func.functional_call over views into the stacked [k, P] tensor —
+    # float buffers like BN running stats are part of the flat state, so each
+    # candidate is evaluated with ITS OWN stats, matching the serial path).
+    # Falls back to the serial loop for models vmap cannot transform.
+
+    _vmap_ok: "bool | None" = None
+
+    def _stacked_call(self, states: Tensor, x: Tensor) -> Tensor:
+        """Logits [k, B, C] of all k candidate states on one input batch."""
+        import torch.func as F
+
+        spec = self.store.spec
+        model = self.store.model
+        batched = {}
+        for e in spec.entries:
+            batched[e.name] = states[:, e.offset : e.offset + e.numel].view(
+                states.shape[0], *e.shape
+            )
+        shared = {
+            name: b
+            for name, b in model.named_buffers()
+            if name not in spec  # non-float buffers (num_batches_tracked)
+        }
+
+        def call_one(params, xin):
+            return F.functional_call(model, {**params, **shared}, (xin,))
+
+        return torch.vmap(call_one, in_dims=(0, None))(batched, x)
+
+    @torch.no_grad()
+    def losses_on_batch(self, states: Tensor, batch) -> Tensor:
+        """CE losses [k] of k stacked flat states on one batch (UBAR stage 2:
+        own + candidates scored in one vmapped forward)."""
+        x, y = batch
+        self.store.model.eval()
+        x = x.to(dtype=self.store.dtype)
+        if self.store.channels_last and x.dim() == 4:
+            x = x.contiguous(memory_format=torch.channels_last)
+        if self._vmap_ok is not False:
+            try:
+                logits = self._stacked_call(states.to(self.store.dtype), x)
+                self._vmap_ok = True
+                k, B, C = logits.shape
+                return torch.nn.functional.cross_entropy(
+                    logits.float().reshape(k * B, C),
+                    y.repeat(k),
+                    reduction="none",
+                ).view(k, B).mean(dim=1)
+            except Exception:
+                self._vmap_ok = False
+        return torch.stack(
+            [self.loss_on_batch(states[i], (x, y)) for i in range(states.shape[0])]
+        )
+
+    @torch.no_grad()
+    def evidential_scores(self, states: Tensor, max_samples: int = 100):
+        """(vacuity [k], accuracy [k]) of k stacked candidate states on local
+        data — one vmapped forward per data batch instead of k model swaps."""
+        if self._vmap_ok is False:
+            pairs = [self.evidential_score(states[i], max_samples)
+                     for i in range(states.shape[0])]
+            return (torch.stack([p[0] for p in pairs]),
+                    torch.stack([p[1] for p in pairs]))
+        self.store.model.eval()
+        k = states.shape[0]
+        st = states.to(self.store.dtype)
+        seen = 0
+        vac_sum = torch.zeros((k,), device=self.device)
+        correct = torch.zeros((k,), device=self.device)
+        try:
+            for x, y in self.loader:
+                if seen >= max_samples:
+                    break
+                x = x.to(self.device, dtype=self.store.dtype)
+                if self.store.channels_last and x.dim() == 4:
+                    x = x.contiguous(memory_format=torch.channels_last)
+                y = y.to(self.device)
+                if seen + x.shape[0] > max_samples:
+                    take = max_samples - seen
+                    x, y = x[:take], y[:take]
+                logits = self._stacked_call(st, x)  # [k, B, C]
+                alpha = torch.nn.functional.softplus(logits.float()) + 1.0
+                S = alpha.sum(dim=-1)
+                vac_sum += (alpha.shape[-1] / S).sum(dim=-1)
+                correct += (alpha.argmax(dim=-1) == y.unsqueeze(0)).float().sum(dim=-1)
+                seen += x.shape[0]
+            self._vmap_ok = True
+        except Exception:
+            self._vmap_ok = False
+            return self.evidential_scores(states, max_samples)
+        n = max(1, seen)
+        return vac_sum / n, correct / n
+
 
 class Aggregator(abc.ABC):
     """Base aggregator. Subclasses implement ``aggregate``; every node gets

@@ -102,20 +102,27 @@ class EvidentialTrustAggregator(Aggregator):
         if neighbor_ids is None:
             neighbor_ids = list(range(k))
 
-        trusts = []
-        for i in range(k):
-            vac, acc = eval_context.evidential_score(
-                neighbor_states[i], self.max_eval_samples
-            )
-            new_trust = self._raw_trust(vac, acc)
-            nid = neighbor_ids[i]
-            if nid in self._trust:
-                new_trust = (
-                    self.gamma_ema * new_trust + (1.0 - self.gamma_ema) * self._trust[nid]
+        # all k neighbors scored in ONE vmapped forward per data batch
+        # (reference: deepcopy + forward per neighbor, evidential_trust.py:237)
+        vac, acc = eval_context.evidential_scores(
+            neighbor_states, self.max_eval_samples
+        )
+        new_trust = self._raw_trust(vac, acc)  # [k]
+        prev = torch.stack(
+            [
+                self._trust.get(nid, torch.zeros((), device=new_trust.device)).to(
+                    new_trust.device
                 )
-            self._trust[nid] = new_trust.detach()
-            trusts.append(new_trust)
-        trust_vec = torch.stack(trusts).float().view(-1)
+                for nid in neighbor_ids
+            ]
+        )
+        has_prev = torch.tensor(
+            [nid in self._trust for nid in neighbor_ids], device=new_trust.device
+        )
+        ema = self.gamma_ema * new_trust + (1.0 - self.gamma_ema) * prev
+        trust_vec = torch.where(has_prev, ema, new_trust).float().view(-1)
+        for i, nid in enumerate(neighbor_ids):
+            self._trust[nid] = trust_vec[i].detach()
 
         tau = self._threshold(round_num)
         accept = trust_vec >= tau

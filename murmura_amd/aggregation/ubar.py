@@ -70,12 +70,14 @@ class UBARAggregator(Aggregator):
         _, keep_idx = torch.topk(dists, num_keep, largest=False)
         candidates = neighbor_states.index_select(0, keep_idx)
 
-        # ---- stage 2: performance filter on one training batch
+        # ---- stage 2: performance filter on one training batch — own +
+        # all candidates scored in ONE vmapped forward (no per-candidate
+        # model swap; reference looped deepcopy+forward, ubar.py:152-222)
         batch = eval_context.next_batch()
-        own_loss = eval_context.loss_on_batch(own_state, batch)
-        cand_losses = torch.stack(
-            [eval_context.loss_on_batch(candidates[i], batch) for i in range(num_keep)]
+        all_losses = eval_context.losses_on_batch(
+            torch.cat([own_state.unsqueeze(0), candidates], dim=0), batch
         )
+        own_loss, cand_losses = all_losses[0], all_losses[1:]
         keep2 = cand_losses <= own_loss
         cnt = keep2.sum()
         fallback = torch.zeros_like(cand_losses)

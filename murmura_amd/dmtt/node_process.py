@@ -61,11 +61,12 @@ class DMTTRoundLoop(FLRoundLoop):
         order = list(received.keys())
         if not order:
             return
-        pairs = [ctx.evidential_score(received[j], max_samples=100) for j in order]
-        flat = torch.stack([t for vac_acc in pairs for t in vac_acc])
-        vals = flat.tolist()  # single host sync for all neighbors
+        vac, acc = ctx.evidential_scores(
+            torch.stack([received[j] for j in order]), max_samples=100
+        )
+        vals = torch.stack([vac, acc]).tolist()  # single host sync
         for idx, j in enumerate(order):
-            self.dmtt_state.record_model_score(j, vals[2 * idx], vals[2 * idx + 1])
+            self.dmtt_state.record_model_score(j, vals[0][idx], vals[1][idx])
 
     def _process_topo_claims(
         self, claims: Dict[int, List[int]], topo, round_num: int

@@ -263,3 +263,30 @@ def test_compat_average_states_copies_nonfloat():
     d = compute_model_distance(s1, s2)
     f1, f2 = flatten_model_state(s1), flatten_model_state(s2)
     assert d == pytest.approx((f1 - f2).norm().item(), rel=1e-5)
+
+
+def test_batched_scoring_matches_serial():
+    """EvalContext.losses_on_batch / evidential_scores (one vmapped forward
+    for all candidates) must equal the per-candidate serial path."""
+    from murmura_amd.aggregation.base import EvalContext
+    from murmura_amd.core.flat import FlatParamStore
+    from murmura_amd.models.evidential import EvidentialHARClassifier
+    from torch.utils.data import DataLoader, TensorDataset
+
+    torch.manual_seed(0)
+    model = EvidentialHARClassifier(input_dim=20, hidden_dims=(16,), num_classes=4)
+    store = FlatParamStore(model, torch.device("cpu"), torch.float32)
+    ds = TensorDataset(torch.randn(64, 20), torch.randint(0, 4, (64,)))
+    ctx = EvalContext(store, DataLoader(ds, batch_size=32), torch.device("cpu"),
+                      evidential=True)
+    base = store.flat.clone()
+    states = torch.stack([base + 0.05 * torch.randn_like(base) for _ in range(3)])
+    batch = ctx.next_batch()
+    batched = ctx.losses_on_batch(states, batch)
+    serial = torch.stack([ctx.loss_on_batch(states[i], batch) for i in range(3)])
+    assert ctx._vmap_ok is True  # the fast path actually ran
+    assert torch.allclose(batched, serial, atol=1e-5)
+    v, a = ctx.evidential_scores(states, 64)
+    pairs = [ctx.evidential_score(states[i], 64) for i in range(3)]
+    assert torch.allclose(v, torch.stack([p[0] for p in pairs]), atol=1e-5)
+    assert torch.allclose(a, torch.stack([p[1] for p in pairs]), atol=1e-5)
