@@ -5,10 +5,13 @@ neighbors, so this is where UBAR/EvidentialTrust candidate scoring shows).
 VERDICT round-1 item 3 'Done' check: aggregate-phase time for ubar /
 evidential_trust comparable to krum's.
 """
+import os
+import sys
 import time
 
 import torch
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from murmura_amd.config.schema import Config
 from murmura_amd.core.network import Network
 from murmura_amd.utils import factories
