@@ -34,6 +34,7 @@ class EvalContext:
         self.device = device
         self.evidential = evidential
         self._batch_iter = None
+        self._eval_cache: "dict[int, tuple]" = {}
 
     def next_batch(self):
         """One training batch, cycling through the loader."""
@@ -46,6 +47,28 @@ class EvalContext:
             batch = next(self._batch_iter)
         x, y = batch
         return x.to(self.device), y.to(self.device)
+
+    def _eval_data(self, max_samples: int):
+        """First ``max_samples`` local samples staged on-device ONCE — the
+        scoring paths run every round per neighbor; re-iterating the CPU
+        DataLoader + H2D per call dominated EvidentialTrust's aggregate
+        phase (measured 3.9 ms/node at m=8 on MI355X)."""
+        hit = self._eval_cache.get(max_samples)
+        if hit is not None:
+            return hit
+        xs, ys, seen = [], [], 0
+        for x, y in self.loader:
+            xs.append(x)
+            ys.append(y)
+            seen += x.shape[0]
+            if seen >= max_samples:
+                break
+        x = torch.cat(xs)[:max_samples].to(self.device, dtype=self.store.dtype)
+        if self.store.channels_last and x.dim() == 4:
+            x = x.contiguous(memory_format=torch.channels_last)
+        y = torch.cat(ys)[:max_samples].to(self.device)
+        self._eval_cache[max_samples] = (x, y)
+        return x, y
 
     @torch.no_grad()
     def loss_on_batch(self, flat_state: Tensor, batch) -> Tensor:
@@ -63,29 +86,15 @@ class EvalContext:
     @torch.no_grad()
     def evidential_score(self, flat_state: Tensor, max_samples: int = 100):
         """(mean vacuity, accuracy) of a foreign evidential model on local
-        data (EvidentialTrust / DMTT model scoring)."""
+        data (EvidentialTrust / DMTT model scoring). Uses the same device-
+        cached eval subset as the batched path so both score identically."""
         self.store.copy_from_flat(flat_state)
         self.store.model.eval()
-        seen = 0
-        vac_sum = torch.zeros((), device=self.device)
-        correct = torch.zeros((), device=self.device)
-        for x, y in self.loader:
-            if seen >= max_samples:
-                break
-            x = x.to(self.device, dtype=self.store.dtype)
-            if self.store.channels_last and x.dim() == 4:
-                x = x.contiguous(memory_format=torch.channels_last)
-            y = y.to(self.device)
-            if seen + x.shape[0] > max_samples:
-                take = max_samples - seen
-                x, y = x[:take], y[:take]
-            logits = self.store.model(x)
-            v, _, _, c = ops.evidential_stats(logits, y)
-            vac_sum = vac_sum + v
-            correct = correct + c
-            seen += x.shape[0]
-        n = max(1, seen)
-        return vac_sum / n, correct.float() / n
+        x, y = self._eval_data(max_samples)
+        logits = self.store.model(x)
+        v, _, _, c = ops.evidential_stats(logits, y)
+        n = max(1, x.shape[0])
+        return v / n, c.float() / n
 
     # ------------------------------------------------------ batched scoring
     # VERDICT round-1 weak #3: UBAR stage-2 / EvidentialTrust / DMTT scoring
@@ -155,34 +164,20 @@ class EvalContext:
             return (torch.stack([p[0] for p in pairs]),
                     torch.stack([p[1] for p in pairs]))
         self.store.model.eval()
-        k = states.shape[0]
         st = states.to(self.store.dtype)
-        seen = 0
-        vac_sum = torch.zeros((k,), device=self.device)
-        correct = torch.zeros((k,), device=self.device)
         try:
-            for x, y in self.loader:
-                if seen >= max_samples:
-                    break
-                x = x.to(self.device, dtype=self.store.dtype)
-                if self.store.channels_last and x.dim() == 4:
-                    x = x.contiguous(memory_format=torch.channels_last)
-                y = y.to(self.device)
-                if seen + x.shape[0] > max_samples:
-                    take = max_samples - seen
-                    x, y = x[:take], y[:take]
-                logits = self._stacked_call(st, x)  # [k, B, C]
-                alpha = torch.nn.functional.softplus(logits.float()) + 1.0
-                S = alpha.sum(dim=-1)
-                vac_sum += (alpha.shape[-1] / S).sum(dim=-1)
-                correct += (alpha.argmax(dim=-1) == y.unsqueeze(0)).float().sum(dim=-1)
-                seen += x.shape[0]
+            x, y = self._eval_data(max_samples)
+            logits = self._stacked_call(st, x)  # [k, n, C]
             self._vmap_ok = True
         except Exception:
             self._vmap_ok = False
             return self.evidential_scores(states, max_samples)
-        n = max(1, seen)
-        return vac_sum / n, correct / n
+        alpha = torch.nn.functional.softplus(logits.float()) + 1.0
+        S = alpha.sum(dim=-1)
+        n = max(1, x.shape[0])
+        vac = (alpha.shape[-1] / S).sum(dim=-1) / n
+        acc = (alpha.argmax(dim=-1) == y.unsqueeze(0)).float().sum(dim=-1) / n
+        return vac, acc
 
 
 class Aggregator(abc.ABC):
