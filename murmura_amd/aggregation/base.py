@@ -36,17 +36,19 @@ class EvalContext:
         self._batch_iter = None
         self._eval_cache: "dict[int, tuple]" = {}
 
+    _device_batches: "list | None" = None
+    _batch_pos: int = 0
+
     def next_batch(self):
-        """One training batch, cycling through the loader."""
-        if self._batch_iter is None:
-            self._batch_iter = iter(self.loader)
-        try:
-            batch = next(self._batch_iter)
-        except StopIteration:
-            self._batch_iter = iter(self.loader)
-            batch = next(self._batch_iter)
-        x, y = batch
-        return x.to(self.device), y.to(self.device)
+        """One training batch, cycling — batches staged on-device once (UBAR
+        calls this every round; per-call H2D was measurable)."""
+        if self._device_batches is None:
+            self._device_batches = [
+                (x.to(self.device), y.to(self.device)) for x, y in self.loader
+            ]
+        batch = self._device_batches[self._batch_pos % len(self._device_batches)]
+        self._batch_pos += 1
+        return batch
 
     def _eval_data(self, max_samples: int):
         """First ``max_samples`` local samples staged on-device ONCE — the
