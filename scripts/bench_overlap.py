@@ -22,14 +22,14 @@ MODEL = {"factory": "models.widemlp",
 # ~34M params fp32 -> 136 MB per state on the wire
 
 
-def _cfg(algo, overlap, sketch_wire, port):
-    return {
+def _cfg(algo, overlap, sketch_wire, port, world=2, attack=False):
+    d = {
         "experiment": {"name": "ovl", "seed": 42, "rounds": 1, "verbose": False},
-        "topology": {"type": "fully", "num_nodes": 2},
+        "topology": {"type": "fully", "num_nodes": world},
         "aggregation": {"algorithm": algo},
         "training": {"local_epochs": 1, "batch_size": 32, "lr": 0.01},
         "data": {"adapter": "synthetic",
-                 "params": {"num_samples": 128, "num_features": 4096,
+                 "params": {"num_samples": 64 * world, "num_features": 4096,
                             "num_classes": 62}},
         "model": MODEL,
         "backend": "distributed",
@@ -38,9 +38,13 @@ def _cfg(algo, overlap, sketch_wire, port):
                         "sketch_wire_mode": sketch_wire},
         "compute": {"dtype": "fp32"},
     }
+    if attack:
+        d["attack"] = {"enabled": True, "type": "gaussian", "percentage": 0.3,
+                       "params": {"noise_std": 50.0}}
+    return d
 
 
-def _worker(rank, cfg_json, port, q):
+def _worker(rank, cfg_json, port, q, world):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["MURMURA_GLOO_CUDA"] = "1"
@@ -51,8 +55,8 @@ def _worker(rank, cfg_json, port, q):
     from murmura_amd.parallel.node_process import FLRoundLoop, init_distributed
 
     cfg = Config(**json.loads(cfg_json))
-    device = init_distributed(cfg, rank, 2)
-    loop = FLRoundLoop(cfg, rank, 2, device)
+    device = init_distributed(cfg, rank, world)
+    loop = FLRoundLoop(cfg, rank, world, device)
     loop.run_round(0)  # warmup: captures, comm setup
     if torch.cuda.is_available():
         torch.cuda.synchronize()
@@ -75,12 +79,12 @@ def _worker(rank, cfg_json, port, q):
     dist.destroy_process_group()
 
 
-def run(algo, overlap, sketch_wire, port):
+def run(algo, overlap, sketch_wire, port, world=2, attack=False):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    cfg = _cfg(algo, overlap, sketch_wire, port)
-    procs = [ctx.Process(target=_worker, args=(r, json.dumps(cfg), port, q))
-             for r in range(2)]
+    cfg = _cfg(algo, overlap, sketch_wire, port, world, attack)
+    procs = [ctx.Process(target=_worker, args=(r, json.dumps(cfg), port, q, world))
+             for r in range(world)]
     for p in procs:
         p.start()
     res = q.get(timeout=300)
@@ -92,14 +96,14 @@ def run(algo, overlap, sketch_wire, port):
 def main():
     port = 29751
     rows = []
-    for algo, overlap, wire, label in [
-        ("krum", False, False, "krum P2P unchunked"),
-        ("krum", True, False, "krum chunked+gram-overlap"),
-        ("fedavg", False, False, "fedavg allreduce fast path"),
-        ("sketchguard", False, False, "sketchguard full exchange"),
-        ("sketchguard", False, True, "sketchguard sketch-wire"),
+    for algo, overlap, wire, world, attack, label in [
+        ("krum", False, False, 2, False, "krum P2P unchunked w2"),
+        ("krum", True, False, 2, False, "krum chunked+gram-overlap w2"),
+        ("fedavg", False, False, 2, False, "fedavg allreduce fast path w2"),
+        ("sketchguard", False, False, 4, True, "sketchguard full exchange w4+atk"),
+        ("sketchguard", False, True, 4, True, "sketchguard sketch-wire w4+atk"),
     ]:
-        r = run(algo, overlap, wire, port)
+        r = run(algo, overlap, wire, port, world, attack)
         port += 1
         rows.append((label, r))
         print(f"{label:32s} {r['ms_per_round']:8.1f} ms/round "
