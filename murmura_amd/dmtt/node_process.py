@@ -16,6 +16,7 @@ kept for the multi-machine/lossy case and for parity with the protocol.
 
 from __future__ import annotations
 
+import os
 from typing import Dict, List, Optional
 
 import torch
@@ -141,3 +142,11 @@ class DMTTRoundLoop(FLRoundLoop):
 
         # 6. select next round's collaborators from G^t direct neighbors
         self._collaborators = self.dmtt_state.top_b(g_neighbors)
+        if os.environ.get("MURMURA_DMTT_DEBUG") == "1" and self.rank == 0:
+            qs = {j: round(self.dmtt_state.collaborator_score(j), 3)
+                  for j in g_neighbors}
+            comp = ([j for j in range(self.world)
+                     if self.attack is not None and self.attack.is_compromised(j)])
+            print(f"[dmtt r{round_num}] rank0 nbrs={g_neighbors} q={qs} "
+                  f"chose={self._collaborators} used={use} compromised={comp}",
+                  flush=True)

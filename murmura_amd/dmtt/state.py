@@ -61,10 +61,16 @@ class DMTTNodeState:
 
     # ---------------------------------------------------------------- model score
     def model_score(self, vacuity: float, accuracy: float) -> float:
+        # a poisoned state can drive the evidential head to inf/NaN alpha
+        # (e.g. sigma-30 Gaussian weights); a NaN score would poison the TopB
+        # sort (NaN comparisons are all False), silently keeping attackers in
+        # C_i^t — treat non-finite evidence as zero trust
+        if not (math.isfinite(vacuity) and math.isfinite(accuracy)):
+            return 0.0
         s = (1.0 - vacuity) * (self.cfg.w_a * accuracy + (1.0 - self.cfg.w_a))
         if vacuity > self.cfg.tau_u:
             s *= math.exp(-self.cfg.eta * (vacuity - self.cfg.tau_u))
-        return s
+        return max(0.0, min(1.0, s))
 
     def record_model_score(self, peer: int, vacuity: float, accuracy: float) -> None:
         self.model_scores[peer] = self.model_score(vacuity, accuracy)
