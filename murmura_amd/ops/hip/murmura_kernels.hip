@@ -1006,6 +1006,181 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ 
   }
 }
 
+// ============================================================== K15 wrw conv
+// 3x3 stride-1 pad-1 NHWC bf16 weight gradient on MFMA (v_mfma_f32_32x32x16_bf16;
+// lane mappings verified by ops/hip/mfma_probe.hip).
+//
+// Why hand-written: MIOpen's chosen igemm wrw kernels use gfx90a-era
+// wt32x32x8 tiles (~15-27 us at shapes whose roofline is 2-6 us) and the
+// atomic variants add 2-3 SubTensorOp launches each (workspace zero +
+// fp32->bf16 cast) — together ~450 us of the flagship's 2.3 ms batch
+// (profiles/r02_steady_state.md).
+//
+// Decomposition: dW[k][ky][kx][c] = sum_{n,y,x} X[n,y+ky-1,x+kx-1,c] *
+// dY[n,y,x,k] = 9 GEMMs A^T B with A = X-shifted [rows x C], B = dY
+// [rows x K], sharing B. Per block: one [64c x 64k] output tile, ALL 9 taps
+// (4 waves x one 32x32 quadrant x 9 accumulators = 144 acc VGPRs,
+// __launch_bounds__(256,1)). X is staged per (image, y-chunk) into a
+// ZERO-PADDED LDS image [ychunk+2][W+2][64c] so every tap's shifted read is
+// in-bounds and border masking costs nothing; dY fragments are loaded once
+// per MFMA step and reused by all 9 taps. Image-range splitting (SP blocks
+// per tile) writes fp32 partial slabs; the LAST-ARRIVING block per tile
+// reduces them in-launch and emits bf16 dW — the agent-scope release/
+// acquire + ticket-counter recipe follows cdna_hip_programming.md §5
+// (split-K reduction) exactly, including the compiler-hazard asm waits.
+// Deterministic (ordered slab reduce — unlike MIOpen's atomic igemm).
+
+typedef __bf16 mfma_bf16x8 __attribute__((ext_vector_type(8)));
+typedef float mfma_f32x16 __attribute__((ext_vector_type(16)));
+
+union WrwU8 {
+  mfma_bf16x8 v;
+  unsigned short u[8];
+};
+
+__global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_kernel(
+    const bf16raw* __restrict__ x,   // [N][H][W][C] (channels_last)
+    const bf16raw* __restrict__ dy,  // [N][H][W][K]
+    float* __restrict__ slab,        // [ct][kt][SP][9][64c][64k]
+    bf16raw* __restrict__ dw,        // [K][3][3][C] (channels_last weight)
+    unsigned* __restrict__ counters, // [ct*kt] zeroed; self-cleaning
+    int N, int H, int W, int C, int K, int SP, int upb, int ychunk) {
+  extern __shared__ float lds[];  // shared symbol across TU kernels
+  unsigned short* ldsu = reinterpret_cast<unsigned short*>(lds);
+  const int xpitch = (W + 2) * 64;
+  unsigned short* ldsx = ldsu;                          // [ychunk+2][W+2][64]
+  unsigned short* ldsy = ldsu + (ychunk + 2) * xpitch;  // [ychunk][W][64]
+
+  const int ct = blockIdx.x, kt = blockIdx.y, z = blockIdx.z;
+  const int c0 = ct * 64, k0 = kt * 64;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lm = lane & 31, half = lane >> 5;
+  const int mc = wave >> 1, nk = wave & 1;  // quadrant: c-half, k-half
+
+  mfma_f32x16 acc[9];
+#pragma unroll
+  for (int t = 0; t < 9; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) acc[t][r] = 0.0f;
+
+  const int ychunks_per_img = H / ychunk;
+  const int units = N * ychunks_per_img;
+  const int u_lo = z * upb;
+  const int u_hi = min(units, u_lo + upb);
+  const int steps = ychunk * W / 16;
+
+  for (int u = u_lo; u < u_hi; ++u) {
+    const int n = u / ychunks_per_img;
+    const int yg = (u % ychunks_per_img) * ychunk;
+    // ---- cooperative LDS fill (2 channels per 4-B load)
+    const int xtotal = (ychunk + 2) * (W + 2) * 32;
+    for (int t = threadIdx.x; t < xtotal; t += 256) {
+      const int cpair = t & 31;
+      const int rest = t >> 5;
+      const int xx = rest % (W + 2), yy = rest / (W + 2);
+      const int gy = yg + yy - 1, gx = xx - 1;
+      unsigned v = 0;
+      if ((unsigned)gy < (unsigned)H && (unsigned)gx < (unsigned)W) {
+        v = *reinterpret_cast<const unsigned*>(
+            x + ((((int64_t)n * H + gy) * W + gx) * C + c0 + cpair * 2));
+      }
+      reinterpret_cast<unsigned*>(ldsx)[t] = v;
+    }
+    const int ytotal = ychunk * W * 32;
+    for (int t = threadIdx.x; t < ytotal; t += 256) {
+      const int cpair = t & 31;
+      const int rest = t >> 5;
+      const int xx = rest % W, yy = rest / W;
+      reinterpret_cast<unsigned*>(ldsy)[t] = *reinterpret_cast<const unsigned*>(
+          dy + ((((int64_t)n * H + yg + yy) * W + xx) * K + k0 + cpair * 2));
+    }
+    __syncthreads();
+    // ---- MFMA over 16-pixel row groups
+    for (int s = 0; s < steps; ++s) {
+      WrwU8 b;
+      int offs[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int p = s * 16 + half * 8 + i;
+        b.u[i] = ldsy[p * 64 + nk * 32 + lm];
+        const int yy = p / W, xx = p % W;
+        offs[i] = ((yy + 1) * (W + 2) + (xx + 1)) * 64 + mc * 32 + lm;
+      }
+#pragma unroll
+      for (int tap = 0; tap < 9; ++tap) {
+        const int doff = ((tap / 3 - 1) * (W + 2) + (tap % 3 - 1)) * 64;
+        WrwU8 a;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) a.u[i] = ldsx[offs[i] + doff];
+        acc[tap] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a.v, b.v, acc[tap], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- write this block's fp32 partial slab (plain coalesced stores)
+  const int tile = ct * gridDim.y + kt;
+  float* myslab = slab + ((int64_t)tile * SP + z) * (9 * 64 * 64);
+#pragma unroll
+  for (int tap = 0; tap < 9; ++tap) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * half + mc * 32;  // c
+      const int col = lm + nk * 32;                                 // k
+      myslab[(tap * 64 + row) * 64 + col] = acc[tap][r];
+    }
+  }
+
+  // ---- last-arriver in-launch reduction (guide §5 split-K recipe)
+  __shared__ unsigned last_flag;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    const unsigned prev = __hip_atomic_fetch_add(
+        &counters[tile], 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    last_flag = (prev == (unsigned)SP - 1) ? 1u : 0u;
+    if (last_flag) {
+      // self-clean for the next call (all SP arrivals already counted)
+      __hip_atomic_store(&counters[tile], 0u, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    }
+  }
+  __syncthreads();
+  if (!last_flag) return;
+  if (threadIdx.x == 0) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  __syncthreads();
+
+  // reduce SP slabs -> bf16 dW, staged per tap through LDS so both the slab
+  // reads (k-contiguous) and the dW writes (c-contiguous) coalesce
+  float* ldsred = lds;  // [64c][64k] fp32 = 16 KB
+  const float* base = slab + (int64_t)tile * SP * (9 * 64 * 64);
+  for (int tap = 0; tap < 9; ++tap) {
+    for (int t = threadIdx.x; t < 64 * 64; t += 256) {
+      float s0 = 0.0f, s1 = 0.0f, s2 = 0.0f, s3 = 0.0f;
+      const float* p = base + tap * 64 * 64 + t;
+      int sp = 0;
+      for (; sp + 4 <= SP; sp += 4) {
+        s0 += p[(int64_t)(sp + 0) * 9 * 64 * 64];
+        s1 += p[(int64_t)(sp + 1) * 9 * 64 * 64];
+        s2 += p[(int64_t)(sp + 2) * 9 * 64 * 64];
+        s3 += p[(int64_t)(sp + 3) * 9 * 64 * 64];
+      }
+      for (; sp < SP; ++sp) s0 += p[(int64_t)sp * 9 * 64 * 64];
+      ldsred[t] = (s0 + s1) + (s2 + s3);
+    }
+    __syncthreads();
+    for (int t = threadIdx.x; t < 64 * 64; t += 256) {
+      const int k = t >> 6, c = t & 63;  // c-contiguous write order
+      from_f(dw[((int64_t)(k0 + k) * 9 + tap) * C + c0 + c], ldsred[c * 64 + k]);
+    }
+    __syncthreads();
+  }
+}
+
 // =================================================================
 // bindings
 // =================================================================
@@ -1549,6 +1724,57 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
   return {dx, dweight, dbias};
 }
 
+Tensor conv3x3s1_wrw(Tensor x, Tensor dy) {
+  TORCH_CHECK(x.is_cuda() && dy.is_cuda() && x.dim() == 4 && dy.dim() == 4,
+              "conv3x3s1_wrw: 4-D cuda tensors");
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 && dy.scalar_type() == at::kBFloat16,
+              "conv3x3s1_wrw: bf16 only");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                  dy.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv3x3s1_wrw: channels_last");
+  const int N = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+            W = (int)x.size(3);
+  const int K = (int)dy.size(1);
+  TORCH_CHECK(dy.size(0) == N && dy.size(2) == H && dy.size(3) == W,
+              "conv3x3s1_wrw: stride-1 same-shape only");
+  TORCH_CHECK(C % 64 == 0 && K % 64 == 0 && C <= 512 && K <= 512,
+              "conv3x3s1_wrw: C,K multiples of 64 (<=512)");
+  TORCH_CHECK(W == 4 || W == 8 || W == 16 || W == 32,
+              "conv3x3s1_wrw: W in {4,8,16,32}");
+  const int ychunk = std::min(H, 8);
+  TORCH_CHECK(H % ychunk == 0, "H not divisible by ychunk");
+  const int ct = C / 64, kt = K / 64;
+  const int units = N * (H / ychunk);
+  int SP = std::max(1, std::min(units, 128 / (ct * kt)));
+  const int upb = (units + SP - 1) / SP;
+  SP = (units + upb - 1) / upb;
+  auto fopt = x.options().dtype(at::kFloat);
+  Tensor slab = at::empty({(int64_t)ct * kt * SP * 9 * 64 * 64}, fopt);
+  Tensor dw = at::empty({K, C, 3, 3},
+                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  // per-process ticket counters (<= 8x8 tiles), zeroed once; the kernel's
+  // last arriver self-cleans its entry for the next call
+  static Tensor counters;
+  if (!counters.defined() || counters.device() != x.device()) {
+    counters = at::zeros({64}, x.options().dtype(at::kInt));
+  }
+  const size_t lds = ((size_t)(ychunk + 2) * (W + 2) * 64 +
+                      (size_t)ychunk * W * 64) * sizeof(unsigned short);
+  static bool attr_set = false;
+  if (!attr_set) {
+    (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_kernel,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              128 * 1024);
+    attr_set = true;
+  }
+  dim3 grid(ct, kt, SP);
+  conv3x3s1_wrw_kernel<<<grid, 256, lds, cur_stream()>>>(
+      (const bf16raw*)x.data_ptr(), (const bf16raw*)dy.data_ptr(),
+      slab.data_ptr<float>(), (bf16raw*)dw.data_ptr(),
+      (unsigned*)counters.data_ptr<int>(), N, H, W, C, K, SP, upb, ychunk);
+  return dw;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -1578,6 +1804,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("count_sketch", &count_sketch, "K4: count-sketch projection");
   m.def("count_sketch_g8", &count_sketch_g8,
         "K4: count-sketch with group-of-8 bins (1 atomic per group per row)");
+  m.def("conv3x3s1_wrw", &conv3x3s1_wrw,
+        "K15: 3x3 s1 p1 NHWC bf16 conv weight gradient (MFMA 32x32x16, "
+        "in-launch split-image reduction)");
   m.def("sgd_step", &sgd_step, "K6: fused p -= lr*g");
   m.def("sgd_step_lrt", &sgd_step_lrt, "K6: fused p -= lr*g, lr from device");
   m.def("gaussian_inject", &gaussian_inject, "K10: x + N(0, sigma^2) (Philox)");
