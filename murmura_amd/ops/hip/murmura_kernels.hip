@@ -1758,8 +1758,12 @@ Tensor conv3x3s1_wrw(Tensor x, Tensor dy) {
   if (!counters.defined() || counters.device() != x.device()) {
     counters = at::zeros({64}, x.options().dtype(at::kInt));
   }
-  const size_t lds = ((size_t)(ychunk + 2) * (W + 2) * 64 +
-                      (size_t)ychunk * W * 64) * sizeof(unsigned short);
+  // the last arriver reuses LDS as a [64][64] fp32 reduction tile (16 KB):
+  // small-W shapes would otherwise allocate less than that
+  const size_t lds = std::max<size_t>(
+      ((size_t)(ychunk + 2) * (W + 2) * 64 + (size_t)ychunk * W * 64) *
+          sizeof(unsigned short),
+      64 * 64 * sizeof(float));
   static bool attr_set = false;
   if (!attr_set) {
     (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_kernel,
