@@ -1042,14 +1042,14 @@ __global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_kernel(
     const bf16raw* __restrict__ x,   // [N][H][W][C] (channels_last)
     const bf16raw* __restrict__ dy,  // [N][H][W][K]
     float* __restrict__ slab,        // [ct][kt][SP][9][64c][64k]
-    bf16raw* __restrict__ dw,        // [K][3][3][C] (channels_last weight)
-    unsigned* __restrict__ counters, // [ct*kt] zeroed; self-cleaning
-    int N, int H, int W, int C, int K, int SP, int upb, int ychunk) {
+    int N, int H, int W, int C, int K, int SP, int upb, int ychunk, int ug) {
   extern __shared__ float lds[];  // shared symbol across TU kernels
   unsigned short* ldsu = reinterpret_cast<unsigned short*>(lds);
   const int xpitch = (W + 2) * 64;
-  unsigned short* ldsx = ldsu;                          // [ychunk+2][W+2][64]
-  unsigned short* ldsy = ldsu + (ychunk + 2) * xpitch;  // [ychunk][W][64]
+  const int ximg = (ychunk + 2) * xpitch;     // one image-slice of the X image
+  const int yimg = ychunk * W * 64;           // one image-slice of the dY image
+  unsigned short* ldsx = ldsu;                // [ug][ychunk+2][W+2][64]
+  unsigned short* ldsy = ldsu + ug * ximg;    // [ug][ychunk][W][64]
 
   const int ct = blockIdx.x, kt = blockIdx.y, z = blockIdx.z;
   const int c0 = ct * 64, k0 = kt * 64;
@@ -1065,62 +1065,104 @@ __global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_kernel(
     for (int r = 0; r < 16; ++r) acc[t][r] = 0.0f;
 
   const int ychunks_per_img = H / ychunk;
-  const int units = N * ychunks_per_img;
+  const int units = (N / ug) * ychunks_per_img;  // unit = (image group, y group)
   const int u_lo = z * upb;
   const int u_hi = min(units, u_lo + upb);
-  const int steps = ychunk * W / 16;
+  const int steps = ychunk * W / 16;  // MFMA steps per image slice
 
   for (int u = u_lo; u < u_hi; ++u) {
-    const int n = u / ychunks_per_img;
+    const int n0 = (u / ychunks_per_img) * ug;
     const int yg = (u % ychunks_per_img) * ychunk;
-    // ---- cooperative LDS fill (2 channels per 4-B load)
-    const int xtotal = (ychunk + 2) * (W + 2) * 32;
-    for (int t = threadIdx.x; t < xtotal; t += 256) {
-      const int cpair = t & 31;
-      const int rest = t >> 5;
-      const int xx = rest % (W + 2), yy = rest / (W + 2);
-      const int gy = yg + yy - 1, gx = xx - 1;
-      unsigned v = 0;
-      if ((unsigned)gy < (unsigned)H && (unsigned)gx < (unsigned)W) {
-        v = *reinterpret_cast<const unsigned*>(
-            x + ((((int64_t)n * H + gy) * W + gx) * C + c0 + cpair * 2));
+    // ---- cooperative LDS fill, two-phase (loads buffered 8 deep so the
+    // ds_writes do not force a vmcnt wait per element; loads are issued
+    // UNCONDITIONALLY from a clamped address and zeroed by select, per the
+    // guide trap: conditional loads de-pipeline into per-element waits)
+    {
+      const int xtotal = ug * (ychunk + 2) * (W + 2) * 32;
+      unsigned buf[8];
+      int tt[8];
+      int nb = 0;
+      for (int t = threadIdx.x; t < xtotal; t += 256) {
+        const int cpair = t & 31;
+        const int rest = t >> 5;
+        const int xx = rest % (W + 2);
+        const int rest2 = rest / (W + 2);
+        const int yy = rest2 % (ychunk + 2);
+        const int img = rest2 / (ychunk + 2);
+        const int gy = yg + yy - 1, gx = xx - 1;
+        const bool valid = (unsigned)gy < (unsigned)H && (unsigned)gx < (unsigned)W;
+        const int64_t off = valid
+            ? ((((int64_t)(n0 + img) * H + gy) * W + gx) * C + c0 + cpair * 2)
+            : (int64_t)0;
+        unsigned v = *reinterpret_cast<const unsigned*>(x + off);
+        buf[nb] = valid ? v : 0u;
+        tt[nb] = t;
+        if (++nb == 8) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            reinterpret_cast<unsigned*>(ldsx)[tt[j]] = buf[j];
+          nb = 0;
+        }
       }
-      reinterpret_cast<unsigned*>(ldsx)[t] = v;
+      for (int j = 0; j < nb; ++j)
+        reinterpret_cast<unsigned*>(ldsx)[tt[j]] = buf[j];
     }
-    const int ytotal = ychunk * W * 32;
-    for (int t = threadIdx.x; t < ytotal; t += 256) {
-      const int cpair = t & 31;
-      const int rest = t >> 5;
-      const int xx = rest % W, yy = rest / W;
-      reinterpret_cast<unsigned*>(ldsy)[t] = *reinterpret_cast<const unsigned*>(
-          dy + ((((int64_t)n * H + yg + yy) * W + xx) * K + k0 + cpair * 2));
+    {
+      const int ytotal = ug * ychunk * W * 32;
+      unsigned buf[8];
+      int tt[8];
+      int nb = 0;
+      for (int t = threadIdx.x; t < ytotal; t += 256) {
+        const int cpair = t & 31;
+        const int rest = t >> 5;
+        const int xx = rest % W;
+        const int rest2 = rest / W;
+        const int yy = rest2 % ychunk;
+        const int img = rest2 / ychunk;
+        buf[nb] = *reinterpret_cast<const unsigned*>(
+            dy + ((((int64_t)(n0 + img) * H + yg + yy) * W + xx) * K + k0 + cpair * 2));
+        tt[nb] = t;
+        if (++nb == 8) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            reinterpret_cast<unsigned*>(ldsy)[tt[j]] = buf[j];
+          nb = 0;
+        }
+      }
+      for (int j = 0; j < nb; ++j)
+        reinterpret_cast<unsigned*>(ldsy)[tt[j]] = buf[j];
     }
     __syncthreads();
-    // ---- MFMA over 16-pixel row groups
-    for (int s = 0; s < steps; ++s) {
-      WrwU8 b;
-      int offs[8];
+    // ---- MFMA over 16-pixel groups, all images of the chunk
+    for (int img = 0; img < ug; ++img) {
+      const unsigned short* xi = ldsx + img * ximg;
+      const unsigned short* yi = ldsy + img * yimg;
+      for (int s = 0; s < steps; ++s) {
+        WrwU8 b;
+        int offs[8];
 #pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        const int p = s * 16 + half * 8 + i;
-        b.u[i] = ldsy[p * 64 + nk * 32 + lm];
-        const int yy = p / W, xx = p % W;
-        offs[i] = ((yy + 1) * (W + 2) + (xx + 1)) * 64 + mc * 32 + lm;
-      }
+        for (int i = 0; i < 8; ++i) {
+          const int p = s * 16 + half * 8 + i;
+          b.u[i] = yi[p * 64 + nk * 32 + lm];
+          const int yy = p / W, xx = p % W;
+          offs[i] = ((yy + 1) * (W + 2) + (xx + 1)) * 64 + mc * 32 + lm;
+        }
 #pragma unroll
-      for (int tap = 0; tap < 9; ++tap) {
-        const int doff = ((tap / 3 - 1) * (W + 2) + (tap % 3 - 1)) * 64;
-        WrwU8 a;
+        for (int tap = 0; tap < 9; ++tap) {
+          const int doff = ((tap / 3 - 1) * (W + 2) + (tap % 3 - 1)) * 64;
+          WrwU8 a;
 #pragma unroll
-        for (int i = 0; i < 8; ++i) a.u[i] = ldsx[offs[i] + doff];
-        acc[tap] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a.v, b.v, acc[tap], 0, 0, 0);
+          for (int i = 0; i < 8; ++i) a.u[i] = xi[offs[i] + doff];
+          acc[tap] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(a.v, b.v, acc[tap], 0, 0, 0);
+        }
       }
     }
     __syncthreads();
   }
 
-  // ---- write this block's fp32 partial slab (plain coalesced stores)
+  // ---- write this block's fp32 partial slab (coalesced; a separate
+  // full-chip reduce kernel combines the SP slabs deterministically)
   const int tile = ct * gridDim.y + kt;
   float* myslab = slab + ((int64_t)tile * SP + z) * (9 * 64 * 64);
 #pragma unroll
@@ -1132,52 +1174,40 @@ __global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_kernel(
       myslab[(tap * 64 + row) * 64 + col] = acc[tap][r];
     }
   }
+}
 
-  // ---- last-arriver in-launch reduction (guide §5 split-K recipe)
-  __shared__ unsigned last_flag;
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    const unsigned prev = __hip_atomic_fetch_add(
-        &counters[tile], 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    last_flag = (prev == (unsigned)SP - 1) ? 1u : 0u;
-    if (last_flag) {
-      // self-clean for the next call (all SP arrivals already counted)
-      __hip_atomic_store(&counters[tile], 0u, __ATOMIC_RELAXED,
-                         __HIP_MEMORY_SCOPE_AGENT);
+// reduce kernel: sum the SP slabs of each tile -> bf16 dW [K][3][3][C].
+// grid = (tiles*9, 4): block handles one (tap-tile, 16-wide k quarter);
+// slab reads are k-contiguous (coalesced), dW writes staged through LDS so
+// they are c-contiguous (channels_last weight layout).
+__global__ void conv3x3_wrw_reduce_kernel(const float* __restrict__ slab,
+                                          bf16raw* __restrict__ dw, int C, int K,
+                                          int SP, int kt_dim) {
+  extern __shared__ float lds[];  // [64c][16k]
+  const int tt = blockIdx.x;      // tile*9 + tap
+  const int tile = tt / 9, tap = tt % 9;
+  const int ct = tile / kt_dim, kt = tile % kt_dim;
+  const int kq = blockIdx.y;      // k quarter
+  const int c0 = ct * 64, k0 = kt * 64 + kq * 16;
+  const float* base = slab + ((int64_t)tile * SP * 9 + tap) * (64 * 64) + kq * 16;
+  for (int t = threadIdx.x; t < 64 * 16; t += blockDim.x) {
+    const int c = t >> 4, k = t & 15;
+    const float* p = base + c * 64 + k;
+    float s0 = 0.0f, s1 = 0.0f, s2 = 0.0f, s3 = 0.0f;
+    int sp = 0;
+    for (; sp + 4 <= SP; sp += 4) {
+      s0 += p[(int64_t)(sp + 0) * 9 * 64 * 64];
+      s1 += p[(int64_t)(sp + 1) * 9 * 64 * 64];
+      s2 += p[(int64_t)(sp + 2) * 9 * 64 * 64];
+      s3 += p[(int64_t)(sp + 3) * 9 * 64 * 64];
     }
+    for (; sp < SP; ++sp) s0 += p[(int64_t)sp * 9 * 64 * 64];
+    lds[c * 16 + k] = (s0 + s1) + (s2 + s3);
   }
   __syncthreads();
-  if (!last_flag) return;
-  if (threadIdx.x == 0) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-  __syncthreads();
-
-  // reduce SP slabs -> bf16 dW, staged per tap through LDS so both the slab
-  // reads (k-contiguous) and the dW writes (c-contiguous) coalesce
-  float* ldsred = lds;  // [64c][64k] fp32 = 16 KB
-  const float* base = slab + (int64_t)tile * SP * (9 * 64 * 64);
-  for (int tap = 0; tap < 9; ++tap) {
-    for (int t = threadIdx.x; t < 64 * 64; t += 256) {
-      float s0 = 0.0f, s1 = 0.0f, s2 = 0.0f, s3 = 0.0f;
-      const float* p = base + tap * 64 * 64 + t;
-      int sp = 0;
-      for (; sp + 4 <= SP; sp += 4) {
-        s0 += p[(int64_t)(sp + 0) * 9 * 64 * 64];
-        s1 += p[(int64_t)(sp + 1) * 9 * 64 * 64];
-        s2 += p[(int64_t)(sp + 2) * 9 * 64 * 64];
-        s3 += p[(int64_t)(sp + 3) * 9 * 64 * 64];
-      }
-      for (; sp < SP; ++sp) s0 += p[(int64_t)sp * 9 * 64 * 64];
-      ldsred[t] = (s0 + s1) + (s2 + s3);
-    }
-    __syncthreads();
-    for (int t = threadIdx.x; t < 64 * 64; t += 256) {
-      const int k = t >> 6, c = t & 63;  // c-contiguous write order
-      from_f(dw[((int64_t)(k0 + k) * 9 + tap) * C + c0 + c], ldsred[c * 64 + k]);
-    }
-    __syncthreads();
+  for (int t = threadIdx.x; t < 64 * 16; t += blockDim.x) {
+    const int k = t >> 6, c = t & 63;  // c-contiguous writes
+    from_f(dw[((int64_t)(k0 + k) * 9 + tap) * C + c0 + c], lds[c * 16 + k]);
   }
 }
 
@@ -1743,8 +1773,20 @@ Tensor conv3x3s1_wrw(Tensor x, Tensor dy) {
               "conv3x3s1_wrw: W in {4,8,16,32}");
   const int ychunk = std::min(H, 8);
   TORCH_CHECK(H % ychunk == 0, "H not divisible by ychunk");
+  // image-group size: pack several small images per LDS chunk so the fill
+  // amortizes over >= ~8 MFMA steps and has enough loads in flight
+  int ug = 1;
+  if (W <= 8) {
+    const size_t ximg = (size_t)(ychunk + 2) * (W + 2) * 64 * 2;
+    const size_t yimg = (size_t)ychunk * W * 64 * 2;
+    while (ug * 2 <= N && (size_t)(ug * 2) * (ximg + yimg) <= 100 * 1024 &&
+           ug * 2 * (ychunk * W / 16) <= 32) {
+      ug *= 2;
+    }
+  }
+  TORCH_CHECK(N % ug == 0, "batch not divisible by image group");
   const int ct = C / 64, kt = K / 64;
-  const int units = N * (H / ychunk);
+  const int units = (N / ug) * (H / ychunk);
   int SP = std::max(1, std::min(units, 128 / (ct * kt)));
   const int upb = (units + SP - 1) / SP;
   SP = (units + upb - 1) / upb;
@@ -1752,30 +1794,23 @@ Tensor conv3x3s1_wrw(Tensor x, Tensor dy) {
   Tensor slab = at::empty({(int64_t)ct * kt * SP * 9 * 64 * 64}, fopt);
   Tensor dw = at::empty({K, C, 3, 3},
                         x.options().memory_format(at::MemoryFormat::ChannelsLast));
-  // per-process ticket counters (<= 8x8 tiles), zeroed once; the kernel's
-  // last arriver self-cleans its entry for the next call
-  static Tensor counters;
-  if (!counters.defined() || counters.device() != x.device()) {
-    counters = at::zeros({64}, x.options().dtype(at::kInt));
-  }
-  // the last arriver reuses LDS as a [64][64] fp32 reduction tile (16 KB):
-  // small-W shapes would otherwise allocate less than that
-  const size_t lds = std::max<size_t>(
-      ((size_t)(ychunk + 2) * (W + 2) * 64 + (size_t)ychunk * W * 64) *
-          sizeof(unsigned short),
-      64 * 64 * sizeof(float));
+  const size_t lds =
+      ((size_t)ug * (ychunk + 2) * (W + 2) * 64 + (size_t)ug * ychunk * W * 64) *
+      sizeof(unsigned short);
   static bool attr_set = false;
   if (!attr_set) {
     (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_kernel,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
-                              128 * 1024);
+                              144 * 1024);
     attr_set = true;
   }
   dim3 grid(ct, kt, SP);
   conv3x3s1_wrw_kernel<<<grid, 256, lds, cur_stream()>>>(
       (const bf16raw*)x.data_ptr(), (const bf16raw*)dy.data_ptr(),
-      slab.data_ptr<float>(), (bf16raw*)dw.data_ptr(),
-      (unsigned*)counters.data_ptr<int>(), N, H, W, C, K, SP, upb, ychunk);
+      slab.data_ptr<float>(), N, H, W, C, K, SP, upb, ychunk, ug);
+  dim3 rgrid(ct * kt * 9, 4);
+  conv3x3_wrw_reduce_kernel<<<rgrid, 256, 64 * 16 * sizeof(float), cur_stream()>>>(
+      slab.data_ptr<float>(), (bf16raw*)dw.data_ptr(), C, K, SP, kt);
   return dw;
 }
 
