@@ -1038,11 +1038,14 @@ union WrwU8 {
   unsigned short u[8];
 };
 
+template <int W>
 __global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_kernel(
     const bf16raw* __restrict__ x,   // [N][H][W][C] (channels_last)
     const bf16raw* __restrict__ dy,  // [N][H][W][K]
     float* __restrict__ slab,        // [ct][kt][SP][9][64c][64k]
-    int N, int H, int W, int C, int K, int SP, int upb, int ychunk, int ug) {
+    int N, int H, int C, int K, int SP, int upb, int ug) {
+  constexpr int ychunk = (W == 4) ? 4 : 8;
+  constexpr int steps = ychunk * W / 16;
   extern __shared__ float lds[];  // shared symbol across TU kernels
   unsigned short* ldsu = reinterpret_cast<unsigned short*>(lds);
   const int xpitch = (W + 2) * 64;
@@ -1068,7 +1071,6 @@ __global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_kernel(
   const int units = (N / ug) * ychunks_per_img;  // unit = (image group, y group)
   const int u_lo = z * upb;
   const int u_hi = min(units, u_lo + upb);
-  const int steps = ychunk * W / 16;  // MFMA steps per image slice
 
   for (int u = u_lo; u < u_hi; ++u) {
     const int n0 = (u / ychunks_per_img) * ug;
@@ -1799,15 +1801,28 @@ Tensor conv3x3s1_wrw(Tensor x, Tensor dy) {
       sizeof(unsigned short);
   static bool attr_set = false;
   if (!attr_set) {
-    (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_kernel,
-                              hipFuncAttributeMaxDynamicSharedMemorySize,
-                              144 * 1024);
+    (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_kernel<4>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_kernel<8>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_kernel<16>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_kernel<32>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
     attr_set = true;
   }
   dim3 grid(ct, kt, SP);
-  conv3x3s1_wrw_kernel<<<grid, 256, lds, cur_stream()>>>(
-      (const bf16raw*)x.data_ptr(), (const bf16raw*)dy.data_ptr(),
-      slab.data_ptr<float>(), N, H, W, C, K, SP, upb, ychunk, ug);
+#define WRW_LAUNCH(WW)                                                          \
+  conv3x3s1_wrw_kernel<WW><<<grid, 256, lds, cur_stream()>>>(                   \
+      (const bf16raw*)x.data_ptr(), (const bf16raw*)dy.data_ptr(),              \
+      slab.data_ptr<float>(), N, H, C, K, SP, upb, ug)
+  switch (W) {
+    case 4: WRW_LAUNCH(4); break;
+    case 8: WRW_LAUNCH(8); break;
+    case 16: WRW_LAUNCH(16); break;
+    case 32: WRW_LAUNCH(32); break;
+  }
+#undef WRW_LAUNCH
   dim3 rgrid(ct * kt * 9, 4);
   conv3x3_wrw_reduce_kernel<<<rgrid, 256, 64 * 16 * sizeof(float), cur_stream()>>>(
       slab.data_ptr<float>(), (bf16raw*)dw.data_ptr(), C, K, SP, kt);
