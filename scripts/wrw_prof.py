@@ -1,3 +1,5 @@
+import sys, os
+sys.path.insert(0, '/root/repo')
 import torch
 from murmura_amd.ops import _load_ext
 ext = _load_ext()
