@@ -20,6 +20,13 @@ def _ext():
 
 
 def _wrw_supported(x: Tensor, weight: Tensor) -> bool:
+    import os
+
+    # opt-in: the v2 kernel is fully correct and MFMA-native but still
+    # 1.2-4x slower than MIOpen's hand-scheduled asm on the large shapes
+    # (measured; see profiles/r02_mfma_wrw.md) — enable to use it anyway
+    if os.environ.get("MURMURA_NATIVE_WRW") != "1":
+        return False
     if not (x.is_cuda and x.dtype == torch.bfloat16):
         return False
     if not x.is_contiguous(memory_format=torch.channels_last):

@@ -383,8 +383,9 @@ def test_conv3x3_wrw_matches_torch(shape):
     assert err.item() < 5e-2, f"rel err {err.item()}"
 
 
-def test_conv3x3_module_end_to_end():
+def test_conv3x3_module_end_to_end(monkeypatch):
     """MurmuraConv3x3 inside autograd: dw and dx both correct."""
+    monkeypatch.setenv("MURMURA_NATIVE_WRW", "1")
     from murmura_amd.ops.fused_conv import MurmuraConv3x3
 
     torch.manual_seed(0)
