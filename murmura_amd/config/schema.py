@@ -126,6 +126,11 @@ class DistributedConfig(_Strict):
     # (sketchguard wire-compression mode; reference kept it latent,
     # sketchguard.py:114-132)
     sketch_wire_mode: bool = False
+    # out-of-band metrics: each rank appends rows to
+    # <metrics_dir>/metrics_rank<k>.jsonl (local write, no collective) and
+    # the in-band gather is skipped — the reference Monitor's passive,
+    # crash-tolerant property (monitor.py:1-15). None = in-band gather.
+    metrics_dir: Optional[str] = None
 
     # ---- ZMQ-era keys (reference: config/schema.py:10-51): accepted for
     # compatibility, IGNORED by the RCCL backend (warned at load time).

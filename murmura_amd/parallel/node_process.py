@@ -430,8 +430,19 @@ def run_node_process(
             start_round = ckpt.restore_rank(loop.node, blob)
             if rank == 0 and blob.get("history"):
                 history = blob["history"]
+    # out-of-band mode: metrics become local file appends (no collective —
+    # the reference Monitor's passive/crash-tolerant property, monitor.py:1-15)
+    writer = None
+    if config.distributed.metrics_dir is not None:
+        from murmura_amd.parallel.monitor import MetricsWriter
+
+        writer = MetricsWriter(config.distributed.metrics_dir, rank)
+
     def _flush(handle) -> None:
         metrics = loop.metrics_from(handle)
+        if writer is not None:
+            writer.write(metrics)
+            return
         rows = exchange.gather_metrics(metrics, dst=0)
         if rank == 0:
             _append_history(history, handle.round_num, rows)
@@ -462,6 +473,17 @@ def run_node_process(
                                       history if rank == 0 else None)
     if pending is not None:
         _flush(pending)
+    if writer is not None:
+        writer.close()
+        # rank 0 assembles the same history schema from the files so the
+        # return contract is identical in both modes
+        if rank == 0:
+            from murmura_amd.parallel.monitor import assemble_history
+
+            exchange.barrier(device)  # let all ranks finish their appends
+            history = assemble_history(config.distributed.metrics_dir, world_size)
+        else:
+            exchange.barrier(device)
     # aggregate per-node aggregator statistics on rank 0 (the distributed
     # analogue of Network.get_node_statistics)
     stats = loop.node.aggregator.get_statistics()
