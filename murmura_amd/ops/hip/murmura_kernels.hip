@@ -1581,7 +1581,7 @@ static int64_t bn_check(const Tensor& x) {
 // tensors, fewer for small ones so the finalize reduction over G stays tiny
 // (~>=32KB of input per block)
 static inline int bn_partials_grid(int64_t R, int C, int esize, int rows_per_iter) {
-  static const int cap = env_int("MURMURA_BN_GRID_CAP", 256);
+  static const int cap = env_int("MURMURA_BN_GRID_CAP", 160);  // swept 96/160/256 on MI355X
   int64_t by_bytes = (R * C * esize) / 32768;
   int64_t g = std::min<int64_t>((R + rows_per_iter - 1) / rows_per_iter,
                                 std::max<int64_t>(1, by_bytes));
