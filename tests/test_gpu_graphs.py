@@ -195,3 +195,57 @@ def test_async_eval_pipeline_in_round_loop():
     metrics.append(loop.metrics_from(pending))
     assert [m["round"] for m in metrics] == [0, 1, 2, 3]
     assert metrics[-1]["accuracy"] > metrics[0]["accuracy"] - 0.2  # learning-ish
+
+
+def test_async_eval_pipelined_matches_sync_eval():
+    """Multi-stream stress (SURVEY §5.2): the side-stream evaluator pipelined
+    with next-round training must produce the SAME metrics as synchronous
+    evaluation of the same states — catches snapshot/stream races (e.g. the
+    round-1 missing reverse sync, ADVICE #1) deterministically."""
+    import os
+
+    import torch.distributed as dist
+
+    from murmura_amd.config.schema import Config
+    from murmura_amd.parallel.node_process import FLRoundLoop, init_distributed
+
+    cfg = Config(**{
+        "experiment": {"name": "race", "seed": 7, "rounds": 6, "verbose": False},
+        "topology": {"type": "fully", "num_nodes": 1},
+        "aggregation": {"algorithm": "fedavg"},
+        "training": {"local_epochs": 1, "batch_size": 32, "lr": 0.05},
+        "data": {"adapter": "synthetic",
+                 "params": {"num_samples": 256, "num_features": 20,
+                            "num_classes": 4}},
+        "model": {"factory": "models.mlp",
+                  "params": {"in_features": 20, "hidden": 32, "num_classes": 4}},
+        "compute": {"dtype": "fp32"},
+    })
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29733")
+    if not dist.is_initialized():
+        dist.init_process_group("nccl" if torch.cuda.is_available() else "gloo",
+                                rank=0, world_size=1)
+    device = torch.device("cuda:0")
+    torch.cuda.set_device(device)
+
+    # run 1: pipelined (async side-stream eval overlapping next training)
+    loop = FLRoundLoop(cfg, 0, 1, device)
+    pend, async_rows = None, []
+    for r in range(6):
+        loop.run_round(r)
+        if pend is not None:
+            async_rows.append(loop.metrics_from(pend))
+        pend = loop.evaluate_round_async(r)
+    async_rows.append(loop.metrics_from(pend))
+
+    # run 2: fully synchronous eval after each round (fresh identical loop)
+    loop2 = FLRoundLoop(cfg, 0, 1, device)
+    sync_rows = []
+    for r in range(6):
+        loop2.run_round(r)
+        sync_rows.append(loop2.evaluate_round(r))
+
+    for a, s in zip(async_rows, sync_rows):
+        assert abs(a["accuracy"] - s["accuracy"]) < 1e-5, (a, s)
+        assert abs(a["loss"] - s["loss"]) < 1e-4, (a, s)
