@@ -1178,6 +1178,139 @@ __global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_kernel(
   }
 }
 
+// v3 (W >= 8): transposed, dx-pre-shifted LDS planes. The v2 kernel's A
+// fragments were 8 scalar ds_read_u16 gathers per tap per step (72/step) —
+// latency-bound at the kernel's 1-wave/SIMD occupancy. Here X is staged as
+// THREE dx-baked zero-bordered planes in [c][y][x] (x-major) layout and dY
+// transposed to [k][y][x], so every A/B fragment is ONE 16-B-aligned
+// ds_read_b128: 10 LDS reads per step instead of 80.
+template <int W>
+__global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_t_kernel(
+    const bf16raw* __restrict__ x,   // [N][H][W][C] (channels_last)
+    const bf16raw* __restrict__ dy,  // [N][H][W][K]
+    float* __restrict__ slab,        // [ct][kt][SP][9][64c][64k]
+    int N, int H, int C, int K, int SP, int upb) {
+  constexpr int ychunk = (W == 32) ? 4 : 8;
+  constexpr int steps = ychunk * W / 16;
+  constexpr int PLANE = (ychunk + 2) * W;  // shorts per (d, c) plane; 8 | PLANE
+  constexpr int YPLANE = ychunk * W;       // shorts per k plane of dY-T
+  extern __shared__ float lds[];
+  unsigned short* ldsu = reinterpret_cast<unsigned short*>(lds);
+  unsigned short* ldsx = ldsu;                  // [3][64][PLANE]
+  unsigned short* ldsy = ldsu + 3 * 64 * PLANE; // [64][YPLANE]
+
+  const int ct = blockIdx.x, kt = blockIdx.y, z = blockIdx.z;
+  const int c0 = ct * 64, k0 = kt * 64;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lm = lane & 31, half = lane >> 5;
+  const int mc = wave >> 1, nk = wave & 1;
+
+  mfma_f32x16 acc[9];
+#pragma unroll
+  for (int t = 0; t < 9; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) acc[t][r] = 0.0f;
+
+  const int ychunks_per_img = H / ychunk;
+  const int units = N * ychunks_per_img;
+  const int u_lo = z * upb;
+  const int u_hi = min(units, u_lo + upb);
+
+  for (int u = u_lo; u < u_hi; ++u) {
+    const int n = u / ychunks_per_img;
+    const int yg = (u % ychunks_per_img) * ychunk;
+    // zero the X planes (x borders are never sourced; y borders come from
+    // OOB-clamped zero loads)
+    {
+      unsigned* z32 = reinterpret_cast<unsigned*>(ldsx);
+      const int nz = 3 * 64 * PLANE / 2;
+      for (int t = threadIdx.x; t < nz; t += 256) z32[t] = 0u;
+    }
+    __syncthreads();
+    // X: src-driven transposed fill with the dx shift baked per copy
+    {
+      const int xtotal = (ychunk + 2) * W * 32;  // 4-B loads (2 channels)
+      for (int t = threadIdx.x; t < xtotal; t += 256) {
+        const int cpair = t & 31;
+        const int rest = t >> 5;
+        const int gx = rest % W, yy = rest / W;
+        const int gy = yg + yy - 1;
+        const bool valid = (unsigned)gy < (unsigned)H;
+        const int64_t off = valid
+            ? ((((int64_t)n * H + gy) * W + gx) * C + c0 + cpair * 2)
+            : (int64_t)0;
+        unsigned v = *reinterpret_cast<const unsigned*>(x + off);
+        if (!valid) v = 0u;
+        const unsigned short lo = (unsigned short)(v & 0xFFFFu);
+        const unsigned short hi = (unsigned short)(v >> 16);
+#pragma unroll
+        for (int d = 0; d < 3; ++d) {
+          const int xd = gx + 1 - d;
+          if ((unsigned)xd < (unsigned)W) {
+            const int base = yy * W + xd;
+            ldsx[(d * 64 + cpair * 2) * PLANE + base] = lo;
+            ldsx[(d * 64 + cpair * 2 + 1) * PLANE + base] = hi;
+          }
+        }
+      }
+    }
+    // dY: transposed fill [k][y][x]
+    {
+      const int ytotal = ychunk * W * 32;
+      for (int t = threadIdx.x; t < ytotal; t += 256) {
+        const int cpair = t & 31;
+        const int rest = t >> 5;
+        const int gx = rest % W, yy = rest / W;
+        unsigned v = *reinterpret_cast<const unsigned*>(
+            dy + ((((int64_t)n * H + yg + yy) * W + gx) * K + k0 + cpair * 2));
+        const int base = yy * W + gx;
+        ldsy[(cpair * 2) * YPLANE + base] = (unsigned short)(v & 0xFFFFu);
+        ldsy[(cpair * 2 + 1) * YPLANE + base] = (unsigned short)(v >> 16);
+      }
+    }
+    __syncthreads();
+    for (int s = 0; s < steps; ++s) {
+      // 16 px of this step: row(s) and x-origins per lane half
+      int yrow, x0;
+      if (W == 32) {
+        yrow = s / 2;
+        x0 = (s % 2) * 16 + half * 8;
+      } else if (W == 16) {
+        yrow = s;
+        x0 = half * 8;
+      } else {  // W == 8: each half covers one full row
+        yrow = s * 2 + half;
+        x0 = 0;
+      }
+      const int kcol = nk * 32 + lm;
+      mfma_bf16x8 b = *reinterpret_cast<const mfma_bf16x8*>(
+          &ldsy[kcol * YPLANE + yrow * W + x0]);
+      const int crow = mc * 32 + lm;
+#pragma unroll
+      for (int tap = 0; tap < 9; ++tap) {
+        const int dy_ = tap / 3 - 1, d = tap % 3;
+        const mfma_bf16x8 a = *reinterpret_cast<const mfma_bf16x8*>(
+            &ldsx[(d * 64 + crow) * PLANE + (yrow + dy_ + 1) * W + x0]);
+        acc[tap] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc[tap], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  const int tile = ct * gridDim.y + kt;
+  float* myslab = slab + ((int64_t)tile * SP + z) * (9 * 64 * 64);
+#pragma unroll
+  for (int tap = 0; tap < 9; ++tap) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * half + mc * 32;
+      const int col = lm + nk * 32;
+      myslab[(tap * 64 + row) * 64 + col] = acc[tap][r];
+    }
+  }
+}
+
 // reduce kernel: sum the SP slabs of each tile -> bf16 dW [K][3][3][C].
 // grid = (tiles*9, 4): block handles one (tap-tile, 16-wide k quarter);
 // slab reads are k-contiguous (coalesced), dW writes staged through LDS so
@@ -1773,12 +1906,13 @@ Tensor conv3x3s1_wrw(Tensor x, Tensor dy) {
               "conv3x3s1_wrw: C,K multiples of 64 (<=512)");
   TORCH_CHECK(W == 4 || W == 8 || W == 16 || W == 32,
               "conv3x3s1_wrw: W in {4,8,16,32}");
-  const int ychunk = std::min(H, 8);
+  // W >= 8 uses the transposed v3 kernel (ychunk 4 for W=32, 8 otherwise);
+  // W == 4 keeps the v2 padded-row kernel with image grouping
+  const bool v3 = W >= 8;
+  const int ychunk = v3 ? (W == 32 ? 4 : 8) : std::min(H, 8);
   TORCH_CHECK(H % ychunk == 0, "H not divisible by ychunk");
-  // image-group size: pack several small images per LDS chunk so the fill
-  // amortizes over >= ~8 MFMA steps and has enough loads in flight
   int ug = 1;
-  if (W <= 8) {
+  if (!v3 && W <= 8) {
     const size_t ximg = (size_t)(ychunk + 2) * (W + 2) * 64 * 2;
     const size_t yimg = (size_t)ychunk * W * 64 * 2;
     while (ug * 2 <= N && (size_t)(ug * 2) * (ximg + yimg) <= 100 * 1024 &&
@@ -1796,9 +1930,11 @@ Tensor conv3x3s1_wrw(Tensor x, Tensor dy) {
   Tensor slab = at::empty({(int64_t)ct * kt * SP * 9 * 64 * 64}, fopt);
   Tensor dw = at::empty({K, C, 3, 3},
                         x.options().memory_format(at::MemoryFormat::ChannelsLast));
-  const size_t lds =
-      ((size_t)ug * (ychunk + 2) * (W + 2) * 64 + (size_t)ug * ychunk * W * 64) *
-      sizeof(unsigned short);
+  const size_t lds = v3
+      ? ((size_t)3 * 64 * (ychunk + 2) * W + (size_t)64 * ychunk * W) *
+            sizeof(unsigned short)
+      : ((size_t)ug * (ychunk + 2) * (W + 2) * 64 +
+         (size_t)ug * ychunk * W * 64) * sizeof(unsigned short);
   static bool attr_set = false;
   if (!attr_set) {
     (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_kernel<4>,
@@ -1809,20 +1945,34 @@ Tensor conv3x3s1_wrw(Tensor x, Tensor dy) {
                               hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
     (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_kernel<32>,
                               hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_t_kernel<8>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_t_kernel<16>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    (void)hipFuncSetAttribute((const void*)conv3x3s1_wrw_t_kernel<32>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
     attr_set = true;
   }
   dim3 grid(ct, kt, SP);
+  if (v3) {
+#define WRW_T_LAUNCH(WW)                                                        \
+    conv3x3s1_wrw_t_kernel<WW><<<grid, 256, lds, cur_stream()>>>(               \
+        (const bf16raw*)x.data_ptr(), (const bf16raw*)dy.data_ptr(),            \
+        slab.data_ptr<float>(), N, H, C, K, SP, upb)
+    switch (W) {
+      case 8: WRW_T_LAUNCH(8); break;
+      case 16: WRW_T_LAUNCH(16); break;
+      case 32: WRW_T_LAUNCH(32); break;
+    }
+#undef WRW_T_LAUNCH
+  } else {
 #define WRW_LAUNCH(WW)                                                          \
   conv3x3s1_wrw_kernel<WW><<<grid, 256, lds, cur_stream()>>>(                   \
       (const bf16raw*)x.data_ptr(), (const bf16raw*)dy.data_ptr(),              \
       slab.data_ptr<float>(), N, H, C, K, SP, upb, ug)
-  switch (W) {
-    case 4: WRW_LAUNCH(4); break;
-    case 8: WRW_LAUNCH(8); break;
-    case 16: WRW_LAUNCH(16); break;
-    case 32: WRW_LAUNCH(32); break;
-  }
+    WRW_LAUNCH(4);
 #undef WRW_LAUNCH
+  }
   dim3 rgrid(ct * kt * 9, 4);
   conv3x3_wrw_reduce_kernel<<<rgrid, 256, 64 * 16 * sizeof(float), cur_stream()>>>(
       slab.data_ptr<float>(), (bf16raw*)dw.data_ptr(), C, K, SP, kt);
