@@ -1192,8 +1192,11 @@ __global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_t_kernel(
     int N, int H, int C, int K, int SP, int upb) {
   constexpr int ychunk = (W == 32) ? 4 : 8;
   constexpr int steps = ychunk * W / 16;
-  constexpr int PLANE = (ychunk + 2) * W;  // shorts per (d, c) plane; 8 | PLANE
-  constexpr int YPLANE = ychunk * W;       // shorts per k plane of dY-T
+  // +8-short pad keeps every fragment 16-B aligned while breaking the
+  // power-of-two dword cycle between lanes' planes (unpadded: 64-way LDS
+  // write conflicts and 32-way b128 read conflicts, measured)
+  constexpr int PLANE = (ychunk + 2) * W + 8;
+  constexpr int YPLANE = ychunk * W + 8;
   extern __shared__ float lds[];
   unsigned short* ldsu = reinterpret_cast<unsigned short*>(lds);
   unsigned short* ldsx = ldsu;                  // [3][64][PLANE]
@@ -1317,16 +1320,16 @@ __global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_t_kernel(
 // they are c-contiguous (channels_last weight layout).
 __global__ void conv3x3_wrw_reduce_kernel(const float* __restrict__ slab,
                                           bf16raw* __restrict__ dw, int C, int K,
-                                          int SP, int kt_dim) {
-  extern __shared__ float lds[];  // [64c][16k]
+                                          int SP, int kt_dim, int kw) {
+  extern __shared__ float lds[];  // [64c][kw]
   const int tt = blockIdx.x;      // tile*9 + tap
   const int tile = tt / 9, tap = tt % 9;
   const int ct = tile / kt_dim, kt = tile % kt_dim;
-  const int kq = blockIdx.y;      // k quarter
-  const int c0 = ct * 64, k0 = kt * 64 + kq * 16;
-  const float* base = slab + ((int64_t)tile * SP * 9 + tap) * (64 * 64) + kq * 16;
-  for (int t = threadIdx.x; t < 64 * 16; t += blockDim.x) {
-    const int c = t >> 4, k = t & 15;
+  const int kq = blockIdx.y;      // k slice of width kw
+  const int c0 = ct * 64, k0 = kt * 64 + kq * kw;
+  const float* base = slab + ((int64_t)tile * SP * 9 + tap) * (64 * 64) + kq * kw;
+  for (int t = threadIdx.x; t < 64 * kw; t += blockDim.x) {
+    const int c = t / kw, k = t % kw;
     const float* p = base + c * 64 + k;
     float s0 = 0.0f, s1 = 0.0f, s2 = 0.0f, s3 = 0.0f;
     int sp = 0;
@@ -1337,12 +1340,12 @@ __global__ void conv3x3_wrw_reduce_kernel(const float* __restrict__ slab,
       s3 += p[(int64_t)(sp + 3) * 9 * 64 * 64];
     }
     for (; sp < SP; ++sp) s0 += p[(int64_t)sp * 9 * 64 * 64];
-    lds[c * 16 + k] = (s0 + s1) + (s2 + s3);
+    lds[c * kw + k] = (s0 + s1) + (s2 + s3);
   }
   __syncthreads();
-  for (int t = threadIdx.x; t < 64 * 16; t += blockDim.x) {
+  for (int t = threadIdx.x; t < 64 * kw; t += blockDim.x) {
     const int k = t >> 6, c = t & 63;  // c-contiguous writes
-    from_f(dw[((int64_t)(k0 + k) * 9 + tap) * C + c0 + c], lds[c * 16 + k]);
+    from_f(dw[((int64_t)(k0 + k) * 9 + tap) * C + c0 + c], lds[c * kw + k]);
   }
 }
 
@@ -1931,8 +1934,8 @@ Tensor conv3x3s1_wrw(Tensor x, Tensor dy) {
   Tensor dw = at::empty({K, C, 3, 3},
                         x.options().memory_format(at::MemoryFormat::ChannelsLast));
   const size_t lds = v3
-      ? ((size_t)3 * 64 * (ychunk + 2) * W + (size_t)64 * ychunk * W) *
-            sizeof(unsigned short)
+      ? ((size_t)3 * 64 * ((ychunk + 2) * W + 8) +
+         (size_t)64 * (ychunk * W + 8)) * sizeof(unsigned short)
       : ((size_t)ug * (ychunk + 2) * (W + 2) * 64 +
          (size_t)ug * ychunk * W * 64) * sizeof(unsigned short);
   static bool attr_set = false;
@@ -1973,9 +1976,11 @@ Tensor conv3x3s1_wrw(Tensor x, Tensor dy) {
     WRW_LAUNCH(4);
 #undef WRW_LAUNCH
   }
-  dim3 rgrid(ct * kt * 9, 4);
-  conv3x3_wrw_reduce_kernel<<<rgrid, 256, 64 * 16 * sizeof(float), cur_stream()>>>(
-      slab.data_ptr<float>(), (bf16raw*)dw.data_ptr(), C, K, SP, kt);
+  int kw = 16;  // k-slice width: more blocks when SP (slab count) is large
+  while (kw > 4 && ct * kt * 9 * (64 / kw) < 128) kw /= 2;
+  dim3 rgrid(ct * kt * 9, 64 / kw);
+  conv3x3_wrw_reduce_kernel<<<rgrid, 256, 64 * kw * sizeof(float), cur_stream()>>>(
+      slab.data_ptr<float>(), (bf16raw*)dw.data_ptr(), C, K, SP, kt, kw);
   return dw;
 }
 
