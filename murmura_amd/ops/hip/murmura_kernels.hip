@@ -1236,15 +1236,9 @@ __global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_t_kernel(
       ldsx[(d * 64 + c) * PLANE + yy * W + xcol] = 0;
     }
     __syncthreads();
-    // X: src-driven transposed fill with the dx shift baked per copy.
-    // Loads are buffered 8 deep BEFORE any LDS write: an interleaved
-    // load-then-scatter loop waits vmcnt(0) per element and serializes the
-    // global loads (measured ~15 us/unit of pure latency).
+    // X: src-driven transposed fill with the dx shift baked per copy
     {
       const int xtotal = (ychunk + 2) * W * 32;  // 4-B loads (2 channels)
-      unsigned buf[8];
-      int tt[8];
-      int nb = 0;
       for (int t = threadIdx.x; t < xtotal; t += 256) {
         const int cpair = t & 31;
         const int rest = t >> 5;
@@ -1255,76 +1249,32 @@ __global__ __launch_bounds__(256, 1) void conv3x3s1_wrw_t_kernel(
             ? ((((int64_t)n * H + gy) * W + gx) * C + c0 + cpair * 2)
             : (int64_t)0;
         unsigned v = *reinterpret_cast<const unsigned*>(x + off);
-        buf[nb] = valid ? v : 0u;
-        tt[nb] = t;
-        if (++nb == 8) {
+        if (!valid) v = 0u;
+        const unsigned short lo = (unsigned short)(v & 0xFFFFu);
+        const unsigned short hi = (unsigned short)(v >> 16);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int cpair2 = tt[j] & 31;
-            const int rest2 = tt[j] >> 5;
-            const int gx2 = rest2 % W, yy2 = rest2 / W;
-            const unsigned short lo = (unsigned short)(buf[j] & 0xFFFFu);
-            const unsigned short hi = (unsigned short)(buf[j] >> 16);
-#pragma unroll
-            for (int d = 0; d < 3; ++d) {
-              const int xd = gx2 + 1 - d;
-              if ((unsigned)xd < (unsigned)W) {
-                const int base = yy2 * W + xd;
-                ldsx[(d * 64 + cpair2 * 2) * PLANE + base] = lo;
-                ldsx[(d * 64 + cpair2 * 2 + 1) * PLANE + base] = hi;
-              }
-            }
-          }
-          nb = 0;
-        }
-      }
-      for (int j = 0; j < nb; ++j) {
-        const int cpair2 = tt[j] & 31;
-        const int rest2 = tt[j] >> 5;
-        const int gx2 = rest2 % W, yy2 = rest2 / W;
-        const unsigned short lo = (unsigned short)(buf[j] & 0xFFFFu);
-        const unsigned short hi = (unsigned short)(buf[j] >> 16);
         for (int d = 0; d < 3; ++d) {
-          const int xd = gx2 + 1 - d;
+          const int xd = gx + 1 - d;
           if ((unsigned)xd < (unsigned)W) {
-            const int base = yy2 * W + xd;
-            ldsx[(d * 64 + cpair2 * 2) * PLANE + base] = lo;
-            ldsx[(d * 64 + cpair2 * 2 + 1) * PLANE + base] = hi;
+            const int base = yy * W + xd;
+            ldsx[(d * 64 + cpair * 2) * PLANE + base] = lo;
+            ldsx[(d * 64 + cpair * 2 + 1) * PLANE + base] = hi;
           }
         }
       }
     }
-    // dY: transposed fill [k][y][x], loads buffered like the X fill
+    // dY: transposed fill [k][y][x]
     {
       const int ytotal = ychunk * W * 32;
-      unsigned buf[8];
-      int tt[8];
-      int nb = 0;
       for (int t = threadIdx.x; t < ytotal; t += 256) {
         const int cpair = t & 31;
         const int rest = t >> 5;
         const int gx = rest % W, yy = rest / W;
-        buf[nb] = *reinterpret_cast<const unsigned*>(
+        unsigned v = *reinterpret_cast<const unsigned*>(
             dy + ((((int64_t)n * H + yg + yy) * W + gx) * K + k0 + cpair * 2));
-        tt[nb] = t;
-        if (++nb == 8) {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int cpair2 = tt[j] & 31;
-            const int rest2 = tt[j] >> 5;
-            const int base = (rest2 / W) * W + (rest2 % W);
-            ldsy[(cpair2 * 2) * YPLANE + base] = (unsigned short)(buf[j] & 0xFFFFu);
-            ldsy[(cpair2 * 2 + 1) * YPLANE + base] = (unsigned short)(buf[j] >> 16);
-          }
-          nb = 0;
-        }
-      }
-      for (int j = 0; j < nb; ++j) {
-        const int cpair2 = tt[j] & 31;
-        const int rest2 = tt[j] >> 5;
-        const int base = (rest2 / W) * W + (rest2 % W);
-        ldsy[(cpair2 * 2) * YPLANE + base] = (unsigned short)(buf[j] & 0xFFFFu);
-        ldsy[(cpair2 * 2 + 1) * YPLANE + base] = (unsigned short)(buf[j] >> 16);
+        const int base = yy * W + gx;
+        ldsy[(cpair * 2) * YPLANE + base] = (unsigned short)(v & 0xFFFFu);
+        ldsy[(cpair * 2 + 1) * YPLANE + base] = (unsigned short)(v >> 16);
       }
     }
     __syncthreads();
