@@ -18,6 +18,7 @@ from murmura_amd.ops.fused_bn import (
     MurmuraBNAddReLU,
     MurmuraBNReLU,
 )
+from murmura_amd.ops.fused_conv import MurmuraConv3x3
 
 
 class SimpleMLP(nn.Module):
@@ -110,9 +111,13 @@ class BasicBlock(nn.Module):
 
     def __init__(self, in_ch: int, out_ch: int, stride: int = 1):
         super().__init__()
-        self.conv1 = nn.Conv2d(in_ch, out_ch, 3, stride=stride, padding=1, bias=False)
+        # stride-1 3x3 convs route their weight gradient through the K15
+        # MFMA kernel on the shapes where it measures faster (ops/fused_conv)
+        self.conv1 = (MurmuraConv3x3(in_ch, out_ch, stride=1) if stride == 1
+                      else nn.Conv2d(in_ch, out_ch, 3, stride=stride, padding=1,
+                                     bias=False))
         self.bn1 = MurmuraBNReLU(out_ch)  # ReLU folded into the BN kernels
-        self.conv2 = nn.Conv2d(out_ch, out_ch, 3, stride=1, padding=1, bias=False)
+        self.conv2 = MurmuraConv3x3(out_ch, out_ch, stride=1)
         # block tail relu(bn2(conv2) + shortcut) fused into one kernel each
         # direction (residual grad comes from the same backward launch)
         self.bn2 = MurmuraBNAddReLU(out_ch)

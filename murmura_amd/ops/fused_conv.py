@@ -22,10 +22,12 @@ def _ext():
 def _wrw_supported(x: Tensor, weight: Tensor) -> bool:
     import os
 
-    # opt-in: the v2 kernel is fully correct and MFMA-native but still
-    # 1.2-4x slower than MIOpen's hand-scheduled asm on the large shapes
-    # (measured; see profiles/r02_mfma_wrw.md) — enable to use it anyway
-    if os.environ.get("MURMURA_NATIVE_WRW") != "1":
+    # MURMURA_NATIVE_WRW: "0" disables entirely, "1" enables all supported
+    # shapes; default = only the shapes where the v3 kernel MEASURES faster
+    # than MIOpen's asm igemm path (W=8: 53.7 vs 61.0 us + no SubTensorOp
+    # launches — profiles/r02_mfma_wrw.md)
+    mode = os.environ.get("MURMURA_NATIVE_WRW", "auto")
+    if mode == "0":
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16):
         return False
@@ -33,8 +35,11 @@ def _wrw_supported(x: Tensor, weight: Tensor) -> bool:
         return False
     n, c, h, w = x.shape
     k = weight.shape[0]
-    return (c % 64 == 0 and k % 64 == 0 and c <= 512 and k <= 512
-            and w in (4, 8, 16, 32) and h % min(h, 8) == 0 and _ext() is not None)
+    ok = (c % 64 == 0 and k % 64 == 0 and c <= 512 and k <= 512
+          and w in (4, 8, 16, 32) and h % min(h, 8) == 0 and _ext() is not None)
+    if mode != "1":
+        ok = ok and w == 8
+    return ok
 
 
 class _Conv3x3Fn(torch.autograd.Function):
