@@ -22,11 +22,13 @@ def _ext():
 def _wrw_supported(x: Tensor, weight: Tensor) -> bool:
     import os
 
-    # MURMURA_NATIVE_WRW: "0" disables entirely, "1" enables all supported
-    # shapes; default = only the shapes where the v3 kernel MEASURES faster
-    # than MIOpen's asm igemm path (W=8: 53.7 vs 61.0 us + no SubTensorOp
-    # launches — profiles/r02_mfma_wrw.md)
-    mode = os.environ.get("MURMURA_NATIVE_WRW", "auto")
+    # MURMURA_NATIVE_WRW: "1" enables all supported shapes, "w8" only W=8
+    # (the shape where the standalone op beats MIOpen: 53.7 vs 61.0 us).
+    # Default OFF: inside the captured flagship epoch the full substitution
+    # measured 13.40 vs 13.96 rounds/s — the separate dx (conv2d_input) +
+    # slab+reduce pipeline costs more than the per-op win recovers
+    # (profiles/r02_mfma_wrw.md).
+    mode = os.environ.get("MURMURA_NATIVE_WRW", "0")
     if mode == "0":
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16):
@@ -37,7 +39,7 @@ def _wrw_supported(x: Tensor, weight: Tensor) -> bool:
     k = weight.shape[0]
     ok = (c % 64 == 0 and k % 64 == 0 and c <= 512 and k <= 512
           and w in (4, 8, 16, 32) and h % min(h, 8) == 0 and _ext() is not None)
-    if mode != "1":
+    if mode == "w8":
         ok = ok and w == 8
     return ok
 
