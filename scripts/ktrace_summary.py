@@ -29,7 +29,7 @@ def main():
     print(f"total window {(t1 - t0) / 1e9:.2f}s, {len(rows)} dispatches")
     # anchor the steady window to the sgd_step dispatches (one per train
     # batch, absent from capture warmup / MIOpen find): last `frac` of them
-    sgd = sorted(r[1] for r in rows if "sgd_step" in r[0])
+    sgd = sorted(r[1] for r in rows if "sgd_step" in r[0] or "multi_tensor_apply" in r[0])
     if sgd:
         n = max(1, int(len(sgd) * frac))
         lo = sgd[-n]
