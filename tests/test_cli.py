@@ -60,3 +60,33 @@ def test_cli_list_all():
     result = runner.invoke(app, ["list-components"])
     assert result.exit_code == 0
     assert "topologies" in result.output and "rccl" in result.output
+
+
+def test_cli_run_distributed_backend(tmp_path):
+    """`murmura run` with backend: distributed spawns N processes via the
+    DistributedRunner and returns the same history schema (gloo on CPU)."""
+    import json
+
+    import yaml
+
+    cfg = {
+        "experiment": {"name": "cli-dist", "seed": 42, "rounds": 2,
+                       "verbose": False},
+        "topology": {"type": "ring", "num_nodes": 2},
+        "aggregation": {"algorithm": "fedavg"},
+        "training": {"local_epochs": 1, "batch_size": 16, "lr": 0.05},
+        "data": {"adapter": "synthetic",
+                 "params": {"num_samples": 80, "num_features": 10,
+                            "num_classes": 3}},
+        "model": {"factory": "models.mlp",
+                  "params": {"in_features": 10, "hidden": 8, "num_classes": 3}},
+        "backend": "distributed",
+        "distributed": {"comm_backend": "gloo", "master_port": 29661},
+    }
+    p = tmp_path / "dist.yaml"
+    p.write_text(yaml.safe_dump(cfg))
+    out = tmp_path / "hist.json"
+    result = runner.invoke(app, ["run", str(p), "--quiet", "--output", str(out)])
+    assert result.exit_code == 0, result.output
+    h = json.loads(out.read_text())
+    assert len(h["round"]) == 2
