@@ -81,6 +81,18 @@ def test_count_sketch(dtype, sketch_size):
     )
 
 
+def test_count_sketch_large_s_fallback():
+    """m*S too large for replicated LDS histograms: the per-row fallback
+    kernel must produce the same sketch."""
+    m, p, S = 2, 200_003, 20_000
+    x = _rand((m, p), torch.float32, seed=11)
+    h, s = ref.make_sketch_tables(p, S, seed=13, device=torch.device("cuda"))
+    out = ops.count_sketch(x, h, s, S)
+    expect = ref.count_sketch(x.float().cpu(), h.cpu(), s.cpu(), S)
+    scale = expect.abs().max().clamp_min(1e-6)
+    assert ((out.cpu() - expect).abs().max() / scale).item() < 1e-4
+
+
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 def test_sgd_step(dtype):
     p = _rand((P_ODD,), dtype, seed=8)
