@@ -183,7 +183,7 @@ class EvalContext:
                 compute()
             torch.cuda.current_stream(st.device).wait_stream(s)
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            with torch.cuda.graph(g, capture_error_mode="thread_local"):
                 out = compute()
             entry = (g, static_s, static_x, static_y, out)
             self._graphs[key] = entry
@@ -252,7 +252,9 @@ class EvalContext:
                 compute()
             torch.cuda.current_stream(st.device).wait_stream(s)
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            # thread_local: other threads' allocator traffic (gc freeing
+            # earlier rounds' tensors) must not poison this capture
+            with torch.cuda.graph(g, capture_error_mode="thread_local"):
                 out_vac, out_acc = compute()
             entry = (g, static, out_vac, out_acc)
             self._graphs[key] = entry
