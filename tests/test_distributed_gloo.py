@@ -233,3 +233,15 @@ def test_out_of_band_metrics(tmp_path):
 
     h2 = assemble_history(mdir, world_size=2)
     assert h2["mean_accuracy"] == h_oob["mean_accuracy"]
+
+
+def test_mobility_krum_chunked_world4():
+    """Dynamic G^t + Krum with the chunked Gram-overlap exchange at world 4:
+    per-round edge sets change while the chunked P2P plan stays symmetric."""
+    cfg = _base_config(4, algo="krum", topo="fully", rounds=3)
+    cfg["mobility"] = {"area_size": 100.0, "comm_range": 60.0,
+                       "max_speed": 10.0, "seed": 3, "ensure_connected": True}
+    cfg["aggregation"]["params"] = {"num_compromised": 0}
+    h = _run_distributed(cfg, 4, 29645)
+    assert len(h["round"]) == 3
+    assert all(0.0 <= a <= 1.0 for a in h["mean_accuracy"])
