@@ -145,3 +145,32 @@ def test_gaussian_inject_stats_and_determinism():
 def test_scale_inject():
     x = torch.randn(50)
     assert torch.allclose(ref.scale_inject(x, -5.0), -5.0 * x)
+
+
+def test_sketch_tables_grouped_structure():
+    """make_sketch_tables: bins constant per aligned group of 8, signs per
+    element, deterministic per seed, and sketch distances stay unbiased
+    enough to preserve orderings (the MI355X-native table layout —
+    ops/reference.py)."""
+    import torch
+
+    from murmura_amd.ops import reference as ref
+
+    P, S = 10_000, 1000
+    h, s = ref.make_sketch_tables(P, S, seed=3, device=torch.device("cpu"))
+    h2, s2 = ref.make_sketch_tables(P, S, seed=3, device=torch.device("cpu"))
+    assert torch.equal(h, h2) and torch.equal(s, s2)  # deterministic
+    ng = P // 8
+    hg = h[: ng * 8].view(ng, 8)
+    assert bool((hg == hg[:, :1]).all())  # group-constant bins
+    # signs vary within groups (per-element)
+    sg = s[: ng * 8].view(ng, 8)
+    assert not bool((sg == sg[:, :1]).all())
+    # distance-ordering sanity: a far vector must sketch farther than a near one
+    base = torch.randn(P)
+    near = base + 0.01 * torch.randn(P)
+    far = base + 10.0 * torch.randn(P)
+    sk = lambda v: ref.count_sketch(v.unsqueeze(0), h, s, S)[0]
+    d_near = (sk(base) - sk(near)).norm()
+    d_far = (sk(base) - sk(far)).norm()
+    assert d_far > 10 * d_near
