@@ -245,3 +245,30 @@ def test_mobility_krum_chunked_world4():
     h = _run_distributed(cfg, 4, 29645)
     assert len(h["round"]) == 3
     assert all(0.0 <= a <= 1.0 for a in h["mean_accuracy"])
+
+
+def test_monitor_partial_round_flush(tmp_path):
+    """assemble_history flushes trailing incomplete rounds with whatever
+    arrived (the reference Monitor's partial flush, monitor.py:125-128)."""
+    import json
+
+    from murmura_amd.parallel.monitor import MetricsWriter, assemble_history
+
+    d = str(tmp_path)
+    w0 = MetricsWriter(d, 0)
+    w1 = MetricsWriter(d, 1)
+    for r in range(2):
+        w0.write({"round": r, "node_id": 0, "accuracy": 0.5 + r * 0.1,
+                  "loss": 1.0, "compromised": False})
+        w1.write({"round": r, "node_id": 1, "accuracy": 0.7, "loss": 0.9,
+                  "compromised": False})
+    # rank 0 got one round further (rank 1 "crashed")
+    w0.write({"round": 2, "node_id": 0, "accuracy": 0.9, "loss": 0.5,
+              "compromised": False})
+    w0.close(); w1.close()
+    # torn tail line must be tolerated
+    with open(f"{d}/metrics_rank0.jsonl", "a") as f:
+        f.write('{"round": 3, "acc')
+    h = assemble_history(d, world_size=2)
+    assert h["round"] == [0, 1, 2]
+    assert abs(h["mean_accuracy"][2] - 0.9) < 1e-9  # partial round: rank 0 only
