@@ -96,3 +96,59 @@ def test_mobility_positions_and_determinism(n, seed, r):
     ta, tb = a.topology_at(r), b.topology_at(r)
     assert ta.edges == tb.edges
     assert all(ta.degree(i) >= 1 for i in range(n))  # ensure_connected default
+
+
+# ------------------------------------------------------- aggregation invariants
+from hypothesis import given, settings, strategies as st
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(2, 9), st.integers(3, 40), st.randoms(use_true_random=False))
+def test_fedavg_permutation_invariant(k, p, rnd):
+    """FedAvg must not depend on neighbor order (gossip arrival order is
+    nondeterministic in the reference's ZMQ transport)."""
+    import torch
+
+    from murmura_amd.aggregation import FedAvgAggregator
+
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    own = torch.randn(p, generator=g)
+    nbrs = torch.randn(k, p, generator=g)
+    perm = torch.randperm(k, generator=g)
+    a = FedAvgAggregator().aggregate(0, own, nbrs)
+    b = FedAvgAggregator().aggregate(0, own, nbrs[perm])
+    assert torch.allclose(a, b, atol=1e-5)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(4, 9), st.integers(3, 40), st.randoms(use_true_random=False))
+def test_krum_selects_a_member(k, p, rnd):
+    """Krum returns one of {own, neighbors} verbatim — never a blend."""
+    import torch
+
+    from murmura_amd.aggregation import KrumAggregator
+
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    own = torch.randn(p, generator=g)
+    nbrs = torch.randn(k, p, generator=g)
+    out = KrumAggregator(num_compromised=1).aggregate(0, own, nbrs)
+    members = [own] + [nbrs[i] for i in range(k)]
+    assert any(torch.equal(out, m) for m in members)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(2, 9), st.integers(3, 40), st.randoms(use_true_random=False))
+def test_balance_output_in_convex_hull_bounds(k, p, rnd):
+    """BALANCE's alpha-blend of own + accepted mean stays inside the
+    coordinate-wise min/max envelope of its inputs."""
+    import torch
+
+    from murmura_amd.aggregation import BALANCEAggregator
+
+    g = torch.Generator().manual_seed(rnd.randint(0, 2**31))
+    own = torch.randn(p, generator=g)
+    nbrs = own.unsqueeze(0) + 0.1 * torch.randn(k, p, generator=g)
+    out = BALANCEAggregator().aggregate(0, own, nbrs, round_num=0)
+    allv = torch.cat([own.unsqueeze(0), nbrs])
+    assert bool((out <= allv.max(dim=0).values + 1e-5).all())
+    assert bool((out >= allv.min(dim=0).values - 1e-5).all())
