@@ -692,6 +692,44 @@ __global__ void evidential_stats_kernel(const T* __restrict__ logits,
 // through one LDS histogram pass, then one global atomic per channel per
 // block. (The first version assigned one scalar channel per thread — 2-byte
 // loads, 190 GB/s; packs restore the streaming roofline.)
+// ticket counter for the fused partials+finalize last-arriver (BN calls are
+// stream-ordered within a process, so one counter suffices; the last block
+// self-cleans it for the next call — zero-initialized at module load)
+__device__ unsigned murmura_bn_ticket = 0u;
+
+// reduce ws[0..G)[{c, C+c}] with 8-way unrolled independent accumulators —
+// the loads have no cross-iteration dependency, so 16 stay in flight per
+// thread and the loop is bandwidth- not latency-bound (a naive serial loop
+// here measured 60 us; this form ~2 us).
+__device__ __forceinline__ void bn_reduce_partials(const float* __restrict__ ws,
+                                                   int G, int C, int c,
+                                                   float& s_out, float& q_out) {
+  constexpr int UR = 8;
+  float s[UR], q[UR];
+#pragma unroll
+  for (int u = 0; u < UR; ++u) s[u] = q[u] = 0.0f;
+  int g = 0;
+  for (; g + UR <= G; g += UR) {
+#pragma unroll
+    for (int u = 0; u < UR; ++u) {
+      s[u] += ws[(int64_t)(g + u) * 2 * C + c];
+      q[u] += ws[(int64_t)(g + u) * 2 * C + C + c];
+    }
+  }
+  for (; g < G; ++g) {
+    s[0] += ws[(int64_t)g * 2 * C + c];
+    q[0] += ws[(int64_t)g * 2 * C + C + c];
+  }
+#pragma unroll
+  for (int u = 1; u < UR; ++u) {
+    s[0] += s[u];
+    q[0] += q[u];
+  }
+  s_out = s[0];
+  q_out = q[0];
+}
+
+
 // Round-2 redesign. Round 1 capped the reduction grid at 64 blocks (global
 // atomics serialized beyond that) which left 192 of 256 CUs idle and ran at
 // ~0.6 TB/s, and every call pre-zeroed a [2,C] workspace (a ~4.8 us
@@ -716,7 +754,21 @@ __global__ void bn_partials_kernel(const T* __restrict__ x,
                                    int64_t R, int C,
                                    const float* __restrict__ mean,
                                    const float* __restrict__ invstd,
-                                   float* __restrict__ ws) {
+                                   float* __restrict__ ws,
+                                   // fused-finalize args (guide §6 G16
+                                   // last-arriver recipe); fwd uses
+                                   // eps/momentum/w/b/coef/saved/running,
+                                   // bwd uses w/gcoef/dweight/dbias
+                                   float eps, float momentum,
+                                   const T* __restrict__ w,
+                                   const T* __restrict__ b,
+                                   float* __restrict__ coef,
+                                   float* __restrict__ save_mean,
+                                   float* __restrict__ save_invstd,
+                                   T* __restrict__ running_mean,
+                                   T* __restrict__ running_var,
+                                   T* __restrict__ dweight,
+                                   T* __restrict__ dbias) {
   constexpr int N = Pack16<T>::N;
   const int ppr = C / N;  // packs per row; caller guarantees divisibility
   const int pk = threadIdx.x % ppr;
@@ -798,43 +850,64 @@ __global__ void bn_partials_kernel(const T* __restrict__ x,
     __syncthreads();
     for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) out[c] = lds[c];
   }
+
+  // ---- fused finalize: the LAST-ARRIVING block reduces the G partials and
+  // computes the per-channel coefficients in the same launch (saves a
+  // ~7 us finalize kernel per BN call; 40 of them per flagship batch).
+  // Publish/observe follows cdna_hip_programming.md §6 Guideline 16: plain
+  // stores -> wait -> barrier -> lane-0 agent release fence -> asm vmcnt
+  // wait (ROCm 7.2 drops the post-wbl2 wait otherwise) -> relaxed ticket;
+  // consumer: acquire fence -> barrier -> plain loads.
+  __shared__ unsigned bn_last;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    const unsigned prev = __hip_atomic_fetch_add(
+        &murmura_bn_ticket, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    bn_last = (prev == gridDim.x - 1) ? 1u : 0u;
+    if (bn_last) {
+      __hip_atomic_store(&murmura_bn_ticket, 0u, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    }
+  }
+  __syncthreads();
+  if (!bn_last) return;
+  if (threadIdx.x == 0) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float s, q;
+    bn_reduce_partials(ws, gridDim.x, C, c, s, q);
+    if (!BWD) {
+      const float m = s / (float)R;
+      const float var = fmaxf(q / (float)R - m * m, 0.0f);
+      const float inv = rsqrtf(var + eps);
+      const float sc = inv * (w ? to_f(w[c]) : 1.0f);
+      coef[c] = sc;
+      coef[C + c] = (b ? to_f(b[c]) : 0.0f) - m * sc;
+      save_mean[c] = m;
+      save_invstd[c] = inv;
+      if (running_mean != nullptr) {
+        from_f(running_mean[c],
+               (1.0f - momentum) * to_f(running_mean[c]) + momentum * m);
+        const float unbiased = R > 1 ? var * (float)R / (float)(R - 1) : var;
+        from_f(running_var[c],
+               (1.0f - momentum) * to_f(running_var[c]) + momentum * unbiased);
+      }
+    } else {
+      coef[c] = (w ? to_f(w[c]) : 1.0f) * invstd[c];  // g_scale
+      coef[C + c] = s / (float)R;                     // g_mean
+      coef[2 * C + c] = q / (float)R;                 // g_proj
+      from_f(dweight[c], q);
+      from_f(dbias[c], s);
+    }
+  }
 }
 
 // finalize (fwd): reduce G partials -> mean/invstd, persist saved stats,
 // EMA-update running stats, precompute scale/shift for the norm pass.
 // Threads own channels; reads of ws[g][c] coalesce across threads.
-// reduce ws[0..G)[{c, C+c}] with 8-way unrolled independent accumulators —
-// the loads have no cross-iteration dependency, so 16 stay in flight per
-// thread and the loop is bandwidth- not latency-bound (a naive serial loop
-// here measured 60 us; this form ~2 us).
-__device__ __forceinline__ void bn_reduce_partials(const float* __restrict__ ws,
-                                                   int G, int C, int c,
-                                                   float& s_out, float& q_out) {
-  constexpr int UR = 8;
-  float s[UR], q[UR];
-#pragma unroll
-  for (int u = 0; u < UR; ++u) s[u] = q[u] = 0.0f;
-  int g = 0;
-  for (; g + UR <= G; g += UR) {
-#pragma unroll
-    for (int u = 0; u < UR; ++u) {
-      s[u] += ws[(int64_t)(g + u) * 2 * C + c];
-      q[u] += ws[(int64_t)(g + u) * 2 * C + C + c];
-    }
-  }
-  for (; g < G; ++g) {
-    s[0] += ws[(int64_t)g * 2 * C + c];
-    q[0] += ws[(int64_t)g * 2 * C + C + c];
-  }
-#pragma unroll
-  for (int u = 1; u < UR; ++u) {
-    s[0] += s[u];
-    q[0] += q[u];
-  }
-  s_out = s[0];
-  q_out = q[0];
-}
-
 template <typename T>
 __global__ void bn_finalize_fwd_kernel(const float* __restrict__ ws, int G,
                                        int64_t R, int C, float eps, float momentum,
@@ -1729,28 +1802,48 @@ static inline int bn_partials_grid(int64_t R, int C, int esize, int rows_per_ite
   return (int)std::min<int64_t>(std::max<int64_t>(g, 1), cap);
 }
 
+struct BnFinalizeArgs {
+  float eps = 0.0f, momentum = 0.0f;
+  const void* w = nullptr;
+  const void* b = nullptr;
+  float* coef = nullptr;
+  float* save_mean = nullptr;
+  float* save_invstd = nullptr;
+  void* running_mean = nullptr;
+  void* running_var = nullptr;
+  void* dweight = nullptr;
+  void* dbias = nullptr;
+};
+
 template <typename elem_t>
 static void bn_partials_launch(const Tensor& x, const Tensor* dy, const Tensor* yout,
                                int64_t R, int C, const float* mean,
-                               const float* invstd, Tensor& ws, int G, bool relu) {
+                               const float* invstd, Tensor& ws, int G, bool relu,
+                               const BnFinalizeArgs& fa) {
   constexpr int N = Pack16<elem_t>::N;
   const int nwaves = BLOCK / WAVE;
   const int ppr = C / N;
   const bool pow2 = (ppr & (ppr - 1)) == 0 && ppr <= WAVE;
   size_t lds = (pow2 ? nwaves * 2 * C : 2 * C) * sizeof(float);
+#define BN_FA                                                                  \
+  fa.eps, fa.momentum, (const elem_t*)fa.w, (const elem_t*)fa.b, fa.coef,      \
+      fa.save_mean, fa.save_invstd, (elem_t*)fa.running_mean,                  \
+      (elem_t*)fa.running_var, (elem_t*)fa.dweight, (elem_t*)fa.dbias
   if (dy == nullptr) {
     bn_partials_kernel<elem_t, false, false><<<G, BLOCK, lds, cur_stream()>>>(
         (const elem_t*)x.data_ptr(), nullptr, nullptr, R, C, nullptr, nullptr,
-        ws.data_ptr<float>());
+        ws.data_ptr<float>(), BN_FA);
   } else if (relu) {
     bn_partials_kernel<elem_t, true, true><<<G, BLOCK, lds, cur_stream()>>>(
         (const elem_t*)x.data_ptr(), (const elem_t*)dy->data_ptr(),
-        (const elem_t*)yout->data_ptr(), R, C, mean, invstd, ws.data_ptr<float>());
+        (const elem_t*)yout->data_ptr(), R, C, mean, invstd, ws.data_ptr<float>(),
+        BN_FA);
   } else {
     bn_partials_kernel<elem_t, true, false><<<G, BLOCK, lds, cur_stream()>>>(
         (const elem_t*)x.data_ptr(), (const elem_t*)dy->data_ptr(), nullptr, R, C,
-        mean, invstd, ws.data_ptr<float>());
+        mean, invstd, ws.data_ptr<float>(), BN_FA);
   }
+#undef BN_FA
 }
 
 std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optional<Tensor> b,
@@ -1778,16 +1871,18 @@ std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optiona
     const int rows_per_iter = std::max(1, BLOCK / (C / N));
     const int G = bn_partials_grid(R, C, (int)sizeof(elem_t), rows_per_iter);
     Tensor ws = at::empty({G, 2 * C}, fopt);  // partials: no pre-zeroing
+    BnFinalizeArgs fa;
+    fa.eps = (float)eps;
+    fa.momentum = (float)momentum;
+    fa.w = w.has_value() ? w->data_ptr() : nullptr;
+    fa.b = b.has_value() ? b->data_ptr() : nullptr;
+    fa.coef = coef.data_ptr<float>();
+    fa.save_mean = sm;
+    fa.save_invstd = sm + C;
+    fa.running_mean = running_mean.has_value() ? running_mean->data_ptr() : nullptr;
+    fa.running_var = running_var.has_value() ? running_var->data_ptr() : nullptr;
     bn_partials_launch<elem_t>(x, nullptr, nullptr, R, C, nullptr, nullptr, ws, G,
-                               false);
-    elem_t* rm = running_mean.has_value() ? (elem_t*)running_mean->data_ptr() : nullptr;
-    elem_t* rv = running_var.has_value() ? (elem_t*)running_var->data_ptr() : nullptr;
-    bn_finalize_fwd_kernel<elem_t>
-        <<<grid_for(C, BLOCK), BLOCK, 0, cur_stream()>>>(
-            ws.data_ptr<float>(), G, R, C, (float)eps, (float)momentum,
-            w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-            b.has_value() ? (const elem_t*)b->data_ptr() : nullptr,
-            coef.data_ptr<float>(), sm, sm + C, rm, rv);
+                               false, fa);
     int blocks = grid_for(R * C / N, BLOCK);
     size_t lds = 2 * C * sizeof(float);
     const elem_t* rp = res.has_value() ? (const elem_t*)res->data_ptr() : nullptr;
@@ -1865,14 +1960,13 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
     const int G = bn_partials_grid(R, C, (int)sizeof(elem_t), rows_per_iter);
     Tensor ws = at::empty({G, 2 * C}, fopt);  // partials: no pre-zeroing
     const Tensor* yp = yout.has_value() ? &*yout : nullptr;
+    BnFinalizeArgs fa;
+    fa.w = w.has_value() ? w->data_ptr() : nullptr;
+    fa.coef = gcoef.data_ptr<float>();
+    fa.dweight = dweight.data_ptr();
+    fa.dbias = dbias.data_ptr();
     bn_partials_launch<elem_t>(x, &dy, yp, R, C, mean.data_ptr<float>(),
-                               invstd.data_ptr<float>(), ws, G, relu);
-    bn_finalize_bwd_kernel<elem_t>
-        <<<grid_for(C, BLOCK), BLOCK, 0, st>>>(
-            ws.data_ptr<float>(), G, R, C, invstd.data_ptr<float>(),
-            w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
-            gcoef.data_ptr<float>(), (elem_t*)dweight.data_ptr(),
-            (elem_t*)dbias.data_ptr());
+                               invstd.data_ptr<float>(), ws, G, relu, fa);
     int blocks = grid_for(R * C / N, BLOCK);
     size_t lds = 5 * C * sizeof(float);
     const elem_t* yp2 = yp ? (const elem_t*)yp->data_ptr() : nullptr;
