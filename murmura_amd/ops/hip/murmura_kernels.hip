@@ -851,6 +851,7 @@ __global__ void bn_partials_kernel(const T* __restrict__ x,
     for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) out[c] = lds[c];
   }
 
+  if (coef == nullptr) return;  // fusion disabled: separate finalize kernel
   // ---- fused finalize: the LAST-ARRIVING block reduces the G partials and
   // computes the per-channel coefficients in the same launch (saves a
   // ~7 us finalize kernel per BN call; 40 of them per flagship batch).
@@ -1871,6 +1872,7 @@ std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optiona
     const int rows_per_iter = std::max(1, BLOCK / (C / N));
     const int G = bn_partials_grid(R, C, (int)sizeof(elem_t), rows_per_iter);
     Tensor ws = at::empty({G, 2 * C}, fopt);  // partials: no pre-zeroing
+    static const bool fused = env_int("MURMURA_BN_FUSED", 1) != 0;
     BnFinalizeArgs fa;
     fa.eps = (float)eps;
     fa.momentum = (float)momentum;
@@ -1881,8 +1883,18 @@ std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optiona
     fa.save_invstd = sm + C;
     fa.running_mean = running_mean.has_value() ? running_mean->data_ptr() : nullptr;
     fa.running_var = running_var.has_value() ? running_var->data_ptr() : nullptr;
+    if (!fused) fa.coef = nullptr;
     bn_partials_launch<elem_t>(x, nullptr, nullptr, R, C, nullptr, nullptr, ws, G,
                                false, fa);
+    if (!fused) {
+      elem_t* rm = running_mean.has_value() ? (elem_t*)running_mean->data_ptr() : nullptr;
+      elem_t* rv = running_var.has_value() ? (elem_t*)running_var->data_ptr() : nullptr;
+      bn_finalize_fwd_kernel<elem_t><<<grid_for(C, BLOCK), BLOCK, 0, cur_stream()>>>(
+          ws.data_ptr<float>(), G, R, C, (float)eps, (float)momentum,
+          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+          b.has_value() ? (const elem_t*)b->data_ptr() : nullptr,
+          coef.data_ptr<float>(), sm, sm + C, rm, rv);
+    }
     int blocks = grid_for(R * C / N, BLOCK);
     size_t lds = 2 * C * sizeof(float);
     const elem_t* rp = res.has_value() ? (const elem_t*)res->data_ptr() : nullptr;
@@ -1960,13 +1972,21 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
     const int G = bn_partials_grid(R, C, (int)sizeof(elem_t), rows_per_iter);
     Tensor ws = at::empty({G, 2 * C}, fopt);  // partials: no pre-zeroing
     const Tensor* yp = yout.has_value() ? &*yout : nullptr;
+    static const bool fused2 = env_int("MURMURA_BN_FUSED", 1) != 0;
     BnFinalizeArgs fa;
     fa.w = w.has_value() ? w->data_ptr() : nullptr;
-    fa.coef = gcoef.data_ptr<float>();
+    fa.coef = fused2 ? gcoef.data_ptr<float>() : nullptr;
     fa.dweight = dweight.data_ptr();
     fa.dbias = dbias.data_ptr();
     bn_partials_launch<elem_t>(x, &dy, yp, R, C, mean.data_ptr<float>(),
                                invstd.data_ptr<float>(), ws, G, relu, fa);
+    if (!fused2) {
+      bn_finalize_bwd_kernel<elem_t><<<grid_for(C, BLOCK), BLOCK, 0, st>>>(
+          ws.data_ptr<float>(), G, R, C, invstd.data_ptr<float>(),
+          w.has_value() ? (const elem_t*)w->data_ptr() : nullptr,
+          gcoef.data_ptr<float>(), (elem_t*)dweight.data_ptr(),
+          (elem_t*)dbias.data_ptr());
+    }
     int blocks = grid_for(R * C / N, BLOCK);
     size_t lds = 5 * C * sizeof(float);
     const elem_t* yp2 = yp ? (const elem_t*)yp->data_ptr() : nullptr;
