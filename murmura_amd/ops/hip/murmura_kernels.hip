@@ -1872,7 +1872,7 @@ std::vector<Tensor> bn_fwd_train(Tensor x, c10::optional<Tensor> w, c10::optiona
     const int rows_per_iter = std::max(1, BLOCK / (C / N));
     const int G = bn_partials_grid(R, C, (int)sizeof(elem_t), rows_per_iter);
     Tensor ws = at::empty({G, 2 * C}, fopt);  // partials: no pre-zeroing
-    static const bool fused = env_int("MURMURA_BN_FUSED", 1) != 0;
+    static const bool fused = env_int("MURMURA_BN_FUSED", 0) != 0;
     BnFinalizeArgs fa;
     fa.eps = (float)eps;
     fa.momentum = (float)momentum;
@@ -1972,7 +1972,7 @@ std::vector<Tensor> bn_bwd(Tensor x, Tensor dy, c10::optional<Tensor> w, Tensor 
     const int G = bn_partials_grid(R, C, (int)sizeof(elem_t), rows_per_iter);
     Tensor ws = at::empty({G, 2 * C}, fopt);  // partials: no pre-zeroing
     const Tensor* yp = yout.has_value() ? &*yout : nullptr;
-    static const bool fused2 = env_int("MURMURA_BN_FUSED", 1) != 0;
+    static const bool fused2 = env_int("MURMURA_BN_FUSED", 0) != 0;
     BnFinalizeArgs fa;
     fa.w = w.has_value() ? w->data_ptr() : nullptr;
     fa.coef = fused2 ? gcoef.data_ptr<float>() : nullptr;
