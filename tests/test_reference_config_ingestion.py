@@ -74,7 +74,9 @@ def _shrink_and_substitute(cfg, num_nodes=4, rounds=1):
 
 
 @pytest.mark.parametrize(
-    "name", ["uci_har_evidential_trust.yaml", "krum_gaussian_20pct.yaml"]
+    "name", ["uci_har_evidential_trust.yaml", "krum_gaussian_20pct.yaml",
+             "pamap2_evidential_trust_gaussian_20pct.yaml",
+             "ppg_dalia_balance_directed_deviation_30pct.yaml"]
 )
 def test_reference_config_runs_one_round(name):
     with warnings.catch_warnings():
