@@ -17,7 +17,6 @@ from __future__ import annotations
 import math
 from typing import Any, Dict, List
 
-import torch
 from torch import Tensor
 
 from murmura_amd import ops

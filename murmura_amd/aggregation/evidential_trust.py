@@ -27,7 +27,7 @@ from typing import Any, Dict, List, Optional
 import torch
 from torch import Tensor
 
-from murmura_amd.aggregation.base import Aggregator, EvalContext, _to_float_list, blend
+from murmura_amd.aggregation.base import Aggregator, EvalContext, _to_float_list
 from murmura_amd import ops
 
 

@@ -21,7 +21,6 @@ Enabled by ``distributed.metrics_dir`` in the config; the in-band gather
 from __future__ import annotations
 
 import json
-import os
 import pathlib
 from typing import Dict, List, Optional
 
