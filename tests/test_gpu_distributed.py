@@ -134,3 +134,56 @@ def test_nccl_colocated_ranks_probe():
         pass  # RCCL permitted it — even better
     else:
         pytest.skip(f"RCCL refused co-located ranks (expected): {outcomes}")
+
+
+def _dmtt_worker(rank, cfg_json, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["MURMURA_GLOO_CUDA"] = "1"
+    os.environ["LOCAL_RANK"] = "0"
+    from murmura_amd.config.schema import Config
+    from murmura_amd.parallel.node_process import run_node_process
+
+    h = run_node_process(Config(**json.loads(cfg_json)), rank, world)
+    if rank == 0:
+        q.put(json.dumps({k: v for k, v in h.items() if k != "node_statistics"}))
+
+
+@pytest.mark.timeout(600)
+def test_dmtt_two_ranks_one_gpu():
+    """DMTT end-to-end on GPU (2 ranks, gloo staging): claims, trust
+    scoring via the captured vmapped forward, TopB selection."""
+    cfg = {
+        "experiment": {"name": "gpu-dmtt", "seed": 42, "rounds": 3,
+                       "verbose": False},
+        "topology": {"type": "fully", "num_nodes": 2},
+        "aggregation": {"algorithm": "fedavg"},
+        "attack": {"enabled": True, "type": "topology_liar", "percentage": 0.5,
+                   "params": {"model_attack_type": "gaussian",
+                              "noise_std": 10.0}},
+        "training": {"local_epochs": 1, "batch_size": 16, "lr": 0.05},
+        "data": {"adapter": "synthetic",
+                 "params": {"num_samples": 160, "num_features": 20,
+                            "num_classes": 4}},
+        "model": {"factory": "examples.wearables.uci_har",
+                  "params": {"input_dim": 20, "hidden_dims": [16],
+                             "num_classes": 4}},
+        "backend": "distributed",
+        "distributed": {"comm_backend": "gloo", "master_port": 29726},
+        "mobility": {"area_size": 100.0, "comm_range": 80.0, "max_speed": 5.0,
+                     "seed": 42, "ensure_connected": True},
+        "dmtt": {"budget_B": 1},
+        "compute": {"dtype": "fp32"},
+    }
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_dmtt_worker,
+                         args=(r, json.dumps(cfg), 2, 29726, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    h = json.loads(q.get(timeout=420))
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert len(h["round"]) == 3
