@@ -297,12 +297,14 @@ class FLRoundLoop:
         nbr_sk = all_sk[_t.tensor(nbr_ids, device=own.device)]
         accept = agg.wire_filter(own_sketch, nbr_sk, round_num)
         # fallback: always fetch the sketch-closest neighbor so the
-        # min_neighbors fallback (balance.py:133-135 semantics) has its state
+        # min_neighbors fallback (balance.py:133-135 semantics) has its
+        # state. The P2P plan needs host integers, so ONE device->host sync
+        # is inherent here; accept-mask and closest-index are combined into
+        # that single transfer (round-1 review flagged the extra .item()).
         dists = (nbr_sk.float() - own_sketch.float().unsqueeze(0)).norm(dim=1)
-        closest = int(dists.argmin().item())
-        accept_idx = [i for i, a in enumerate(accept.tolist()) if a]
-        if closest not in accept_idx:
-            accept_idx.append(closest)
+        fetch = accept.clone()
+        fetch[dists.argmin()] = True
+        accept_idx = [i for i, a in enumerate(fetch.tolist()) if a]
         want = [nbr_ids[i] for i in accept_idx]
         self.last_wire_stats = {
             "accepted": len(accept_idx),
