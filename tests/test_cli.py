@@ -90,3 +90,23 @@ def test_cli_run_distributed_backend(tmp_path):
     assert result.exit_code == 0, result.output
     h = json.loads(out.read_text())
     assert len(h["round"]) == 2
+
+
+def test_cli_device_override(tmp_path):
+    import yaml
+
+    cfg = {
+        "experiment": {"name": "dev", "seed": 1, "rounds": 1, "verbose": False},
+        "topology": {"type": "ring", "num_nodes": 2},
+        "aggregation": {"algorithm": "fedavg"},
+        "training": {"batch_size": 8, "lr": 0.05},
+        "data": {"adapter": "synthetic",
+                 "params": {"num_samples": 48, "num_features": 6,
+                            "num_classes": 2}},
+        "model": {"factory": "models.mlp",
+                  "params": {"in_features": 6, "hidden": 8, "num_classes": 2}},
+    }
+    p = tmp_path / "c.yaml"
+    p.write_text(yaml.safe_dump(cfg))
+    result = runner.invoke(app, ["run", str(p), "--quiet", "--device", "cpu"])
+    assert result.exit_code == 0, result.output
