@@ -26,7 +26,7 @@ from murmura_amd.aggregation.sketchguard import SketchguardAggregator
 from murmura_amd.aggregation.ubar import UBARAggregator
 from murmura_amd.aggregation.evidential_trust import EvidentialTrustAggregator
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 __all__ = [
     "Config",
