@@ -81,3 +81,27 @@ def test_unknown_extension(tmp_path):
         load_config(p)
     with pytest.raises(FileNotFoundError):
         load_config(tmp_path / "missing.yaml")
+
+
+def test_save_load_roundtrip_with_legacy_keys(tmp_path):
+    """A config carrying ZMQ-era keys survives save/load (the compat keys
+    are plain fields, so round-tripping preserves them)."""
+    import warnings
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        c = Config(**{
+            "experiment": {"name": "rt", "rounds": 3},
+            "topology": {"type": "ring", "num_nodes": 4},
+            "aggregation": {"algorithm": "krum", "params": {"f": 1}},
+            "training": {}, "data": {}, "model": {},
+            "distributed": {"transport": "tcp", "startup_grace_s": 9.0,
+                            "round_duration_s": 30.0},
+        })
+        p = tmp_path / "rt.yaml"
+        save_config(c, p)
+        c2 = load_config(p)
+    assert c2.distributed.transport == "tcp"
+    assert c2.distributed.startup_grace_s == 9.0
+    assert c2.distributed.round_duration_s == 30.0
+    assert c2.aggregation.params == {"f": 1}
